@@ -1,0 +1,298 @@
+"""Convolutional blocks, CNN stacks and the U-Net used between induced points.
+
+Parity with /root/reference/npf/architectures/cnn.py (GaussianConv2d :24-53,
+ConvBlock :56-123, ResConvBlock :126-215, ResNormalizedConvBlock :218-304,
+CNN :307-380, UnetCNN :383-516).
+
+The depthwise+pointwise (separable) blocks on [Z*B, r_dim, n_induced] /
+[Z*B, r_dim, H, W] are the ConvNP hot loop (SURVEY.md §2.3 "ResConvBlock
+stack"); they run through MIOpen on ROCm, with channel-last permutes kept at
+the CNN boundary exactly as the reference does (cnn.py:363-375).
+"""
+
+import torch
+import torch.nn as nn
+from torch.nn import functional as F
+
+from npf.utils.helpers import (
+    channels_to_2nd_dim,
+    channels_to_last_dim,
+    make_depth_sep_conv,
+)
+from npf.utils.initialization import init_param_, weights_init
+
+__all__ = [
+    "GaussianConv2d",
+    "ConvBlock",
+    "ResNormalizedConvBlock",
+    "ResConvBlock",
+    "CNN",
+    "UnetCNN",
+]
+
+
+class GaussianConv2d(nn.Module):
+    """Separable Gaussian blur with learned per-axis widths
+    (reference cnn.py:24-53)."""
+
+    def __init__(self, kernel_size=5, **kwargs):
+        super().__init__()
+        self.kwargs = kwargs
+        assert kernel_size % 2 == 1
+        self.kernel_sizes = (kernel_size, kernel_size)
+        self.exponent = -(
+            (torch.arange(0, kernel_size).view(-1, 1).float() - kernel_size // 2) ** 2
+        )
+        self.reset_parameters()
+
+    def reset_parameters(self):
+        self.weights_x = nn.Parameter(torch.tensor([1.0]))
+        self.weights_y = nn.Parameter(torch.tensor([1.0]))
+
+    def forward(self, X):
+        self.exponent = self.exponent.to(X.device)
+        marginal_x = torch.softmax(self.exponent * self.weights_x, dim=0)
+        marginal_y = torch.softmax(self.exponent * self.weights_y, dim=0).T
+        in_chan = X.size(1)
+        filters = (marginal_x @ marginal_y).view(1, 1, *self.kernel_sizes)
+        filters = filters.expand(in_chan, 1, *self.kernel_sizes)
+        return F.conv2d(X, filters, groups=in_chan, **self.kwargs)
+
+
+class ConvBlock(nn.Module):
+    """norm -> act -> depthwise-separable conv (reference cnn.py:56-123)."""
+
+    def __init__(
+        self,
+        in_chan,
+        out_chan,
+        Conv,
+        kernel_size=5,
+        dilation=1,
+        activation=nn.ReLU(),
+        Normalization=nn.Identity,
+        **kwargs,
+    ):
+        super().__init__()
+        self.activation = activation
+        Conv = make_depth_sep_conv(Conv)
+        self.conv = Conv(in_chan, out_chan, kernel_size, padding=kernel_size // 2, **kwargs)
+        self.norm = Normalization(in_chan)
+        self.reset_parameters()
+
+    def reset_parameters(self):
+        weights_init(self)
+
+    def forward(self, X):
+        return self.conv(self.activation(self.norm(X)))
+
+
+class ResConvBlock(nn.Module):
+    """Pre-activation residual block with depthwise-separable convs
+    (reference cnn.py:126-215).
+
+    Attribute names (`conv1.depthwise`, `conv2_depthwise`, `conv2_pointwise`,
+    `norm1`, `norm2`) are checkpoint-format-relevant.
+    """
+
+    def __init__(
+        self,
+        in_chan,
+        out_chan,
+        Conv,
+        kernel_size=5,
+        activation=nn.ReLU(),
+        Normalization=nn.Identity,
+        is_bias=True,
+        n_conv_layers=1,
+    ):
+        super().__init__()
+        self.activation = activation
+        self.n_conv_layers = n_conv_layers
+        assert self.n_conv_layers in (1, 2)
+        if kernel_size % 2 == 0:
+            raise ValueError(f"`kernel_size={kernel_size}`, but should be odd.")
+        padding = kernel_size // 2
+
+        if self.n_conv_layers == 2:
+            self.norm1 = Normalization(in_chan)
+            self.conv1 = make_depth_sep_conv(Conv)(
+                in_chan, in_chan, kernel_size, padding=padding, bias=is_bias
+            )
+        self.norm2 = Normalization(in_chan)
+        self.conv2_depthwise = Conv(
+            in_chan, in_chan, kernel_size, padding=padding, groups=in_chan, bias=is_bias
+        )
+        self.conv2_pointwise = Conv(in_chan, out_chan, 1, bias=is_bias)
+        self.reset_parameters()
+
+    def reset_parameters(self):
+        weights_init(self)
+
+    def forward(self, X):
+        out = self.conv1(self.activation(self.norm1(X))) if self.n_conv_layers == 2 else X
+        out = self.conv2_depthwise(self.activation(self.norm2(out)))
+        # residual added BEFORE the pointwise so out_chan may differ
+        out = out + X
+        return self.conv2_pointwise(out.contiguous())
+
+
+class ResNormalizedConvBlock(ResConvBlock):
+    """Residual block with normalized convolutions and confidence channels
+    (reference cnn.py:218-304)."""
+
+    def __init__(
+        self, in_chan, out_chan, Conv, kernel_size=5, activation=nn.ReLU(),
+        is_bias=True, **kwargs,
+    ):
+        super().__init__(
+            in_chan,
+            out_chan,
+            Conv,
+            kernel_size=kernel_size,
+            activation=activation,
+            is_bias=is_bias,
+            Normalization=nn.Identity,
+            **kwargs,
+        )
+
+    def reset_parameters(self):
+        weights_init(self)
+        self.bias = nn.Parameter(torch.tensor([0.0]))
+        self.temperature = nn.Parameter(torch.tensor([0.0]))
+        init_param_(self.temperature)
+
+    def forward(self, X):
+        """First half of channels = signal, second half = confidence."""
+        signal, conf_1 = X.chunk(2, dim=1)
+        conf_1 = conf_1.clamp(min=0, max=1)
+        X = signal * conf_1
+
+        numerator = self.conv1(self.activation(X))
+        numerator = self.conv2_depthwise(self.activation(numerator))
+        density = self.conv2_depthwise(self.conv1(conf_1))
+        out = numerator / torch.clamp(density, min=1e-5)
+
+        # confidence can only grow, capped at 1
+        conf_2 = conf_1 + torch.sigmoid(density * F.softplus(self.temperature) + self.bias)
+        conf_2 = conf_2.clamp(max=1)
+        out = out + X
+
+        out = self.conv2_pointwise(out)
+        conf_2 = self.conv2_pointwise(conf_2)
+        return torch.cat([out, conf_2], dim=1)
+
+
+class CNN(nn.Module):
+    """Stack of conv blocks with optional channel-last I/O
+    (reference cnn.py:307-380)."""
+
+    def __init__(self, n_channels, ConvBlock, n_blocks=3, is_chan_last=False, **kwargs):
+        super().__init__()
+        self.n_blocks = n_blocks
+        self.is_chan_last = is_chan_last
+        self.in_out_channels = self._get_in_out_channels(n_channels, n_blocks)
+        self.conv_blocks = nn.ModuleList(
+            ConvBlock(i, o, **kwargs) for i, o in self.in_out_channels
+        )
+        self.is_return_rep = False
+        self.reset_parameters()
+
+    def reset_parameters(self):
+        weights_init(self)
+
+    def _get_in_out_channels(self, n_channels, n_blocks):
+        if isinstance(n_channels, int):
+            channel_list = [n_channels] * (n_blocks + 1)
+        else:
+            channel_list = list(n_channels)
+        assert len(channel_list) == n_blocks + 1, f"{len(channel_list)} != {n_blocks + 1}"
+        return list(zip(channel_list, channel_list[1:]))
+
+    def forward(self, X):
+        if self.is_chan_last:
+            X = channels_to_2nd_dim(X)
+        X, representation = self.apply_convs(X)
+        if self.is_chan_last:
+            X = channels_to_last_dim(X)
+        if self.is_return_rep:
+            return X, representation
+        return X
+
+    def apply_convs(self, X):
+        for conv_block in self.conv_blocks:
+            X = conv_block(X)
+        return X, None
+
+
+class UnetCNN(CNN):
+    """U-Net over induced points (reference cnn.py:383-516)."""
+
+    def __init__(
+        self,
+        n_channels,
+        ConvBlock,
+        Pool,
+        upsample_mode,
+        max_nchannels=256,
+        pooling_size=2,
+        is_force_same_bottleneck=False,
+        is_return_rep=False,
+        **kwargs,
+    ):
+        self.max_nchannels = max_nchannels
+        super().__init__(n_channels, ConvBlock, **kwargs)
+        self.pooling_size = pooling_size
+        self.pooling = Pool(self.pooling_size)
+        self.upsample_mode = upsample_mode
+        self.is_force_same_bottleneck = is_force_same_bottleneck
+        self.is_return_rep = is_return_rep
+
+    def apply_convs(self, X):
+        n_down_blocks = self.n_blocks // 2
+        residuals = [None] * n_down_blocks
+
+        for i in range(n_down_blocks):
+            X = self.conv_blocks[i](X)
+            residuals[i] = X
+            X = self.pooling(X)
+
+        X = self.conv_blocks[n_down_blocks](X)
+        # global-mean-pool bottleneck summary (reference cnn.py:464)
+        representation = X.view(*X.shape[:2], -1).mean(-1)
+
+        if self.is_force_same_bottleneck and self.training:
+            # average bottlenecks of the two halves of the batch (the halves
+            # hold different context/target draws of the SAME functions)
+            batch_size = X.size(0)
+            X_mean = (X[: batch_size // 2] + X[batch_size // 2 :]) / 2
+            X = torch.cat([X_mean, X_mean], dim=0)
+
+        for i in range(n_down_blocks + 1, self.n_blocks):
+            X = F.interpolate(
+                X,
+                mode=self.upsample_mode,
+                scale_factor=self.pooling_size,
+                align_corners=True,
+            )
+            X = torch.cat((X, residuals[n_down_blocks - i]), dim=1)
+            X = self.conv_blocks[i](X)
+
+        return X, representation
+
+    def _get_in_out_channels(self, n_channels, n_blocks):
+        """Channel-doubling U-Net schedule, capped at `max_nchannels`, with
+        the up-path inputs doubled by the skip concats (reference :492-516)."""
+        factor_chan = 2
+        assert n_blocks % 2 == 1, f"n_blocks={n_blocks} not odd"
+        channel_list = [factor_chan**i * n_channels for i in range(n_blocks // 2 + 1)]
+        channel_list = channel_list + channel_list[::-1]
+        channel_list = (
+            channel_list[:1]
+            + [min(c, self.max_nchannels) for c in channel_list[1:-1]]
+            + channel_list[-1:]
+        )
+        in_out_channels = super()._get_in_out_channels(channel_list, n_blocks)
+        idcs = slice(len(in_out_channels) // 2 + 1, len(in_out_channels))
+        in_out_channels[idcs] = [(i * 2, o) for i, o in in_out_channels[idcs]]
+        return in_out_channels

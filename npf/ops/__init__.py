@@ -1,0 +1,13 @@
+"""MI355X ops layer: fused CDNA4 HIP kernels with torch reference fallbacks."""
+
+from ._backend import extension, has_extension, require_extension
+from .functional import attention_qkv, gaussian_nll_sum, setconv_gaussian
+
+__all__ = [
+    "attention_qkv",
+    "setconv_gaussian",
+    "gaussian_nll_sum",
+    "extension",
+    "has_extension",
+    "require_extension",
+]
