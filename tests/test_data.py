@@ -1,0 +1,169 @@
+"""Data-layer tests: kernels vs sklearn, GPDataset semantics, splitters,
+collate ABI."""
+
+import numpy as np
+import pytest
+import torch
+
+from npf.data import GPDataset, cntxt_trgt_collate
+from npf.data.kernels import (
+    RBF,
+    ExpSineSquared,
+    Matern,
+    WhiteKernel,
+    from_sklearn,
+)
+from npf.utils.datasplit import (
+    CntxtTrgtGetter,
+    GetRandomIndcs,
+    GridCntxtTrgtGetter,
+    RandomMasker,
+    get_all_indcs,
+    no_masker,
+)
+
+
+class TestKernels:
+    @pytest.mark.parametrize(
+        "mine,sk_name,sk_kwargs",
+        [
+            (RBF(0.2), "RBF", dict(length_scale=0.2)),
+            (ExpSineSquared(0.5, 0.5), "ExpSineSquared",
+             dict(length_scale=0.5, periodicity=0.5)),
+            (Matern(0.2, nu=1.5), "Matern", dict(length_scale=0.2, nu=1.5)),
+            (WhiteKernel(0.1), "WhiteKernel", dict(noise_level=0.1)),
+        ],
+    )
+    def test_covariance_matches_sklearn(self, mine, sk_name, sk_kwargs):
+        import sklearn.gaussian_process.kernels as SK
+
+        X = np.sort(np.random.RandomState(0).uniform(-2, 2, (32, 1)), axis=0)
+        K_sk = getattr(SK, sk_name)(**sk_kwargs)(X)
+        K_me = mine(torch.from_numpy(X)).numpy()
+        assert np.allclose(K_me, K_sk, atol=1e-6)
+
+    def test_from_sklearn_composite(self):
+        import sklearn.gaussian_process.kernels as SK
+
+        sk = SK.WhiteKernel(noise_level=0.1) + SK.Matern(length_scale=0.2, nu=1.5)
+        mine = from_sklearn(sk)
+        X = np.random.RandomState(1).uniform(-2, 2, (16, 1))
+        assert np.allclose(mine(torch.from_numpy(X)).numpy(), sk(X), atol=1e-6)
+
+
+class TestGPDataset:
+    def test_shapes_and_range(self):
+        ds = GPDataset(kernel=RBF(0.2), n_samples=64, n_points=32)
+        x, y = ds[0]
+        assert x.shape == (32, 1) and y.shape == (32, 1)
+        assert x.min() >= -1 and x.max() <= 1
+        assert len(ds) == 64
+
+    def test_fresh_epochs_regenerate(self):
+        ds = GPDataset(
+            kernel=RBF(0.2), n_samples=8, n_points=16, is_reuse_across_epochs=False
+        )
+        first = [ds[i][1].clone() for i in range(8)]  # consumes the chunk
+        second = [ds[i][1].clone() for i in range(8)]
+        assert not all(torch.equal(a, b) for a, b in zip(first, second))
+
+    def test_sample_statistics(self):
+        """Marginal variance of an RBF prior draw is ~1."""
+        ds = GPDataset(kernel=RBF(0.2), n_samples=512, n_points=64)
+        var = ds.targets.var().item()
+        assert 0.7 < var < 1.3, var
+
+    def test_chunk_cache_roundtrip(self, tmp_path):
+        f = str(tmp_path / "cache.npz")
+        ds = GPDataset(kernel=RBF(0.2), n_samples=16, n_points=8,
+                       save_file=(f, "rbf"))
+        d0, t0 = ds.data.clone(), ds.targets.clone()
+        ds2 = GPDataset(kernel=RBF(0.2), n_samples=16, n_points=8,
+                        save_file=(f, "rbf"))
+        assert torch.equal(ds2.data, d0) and torch.equal(ds2.targets, t0)
+
+    def test_vary_hyperparameters(self):
+        ds = GPDataset(
+            kernel=Matern(length_scale_bounds=(0.01, 0.3), nu=1.5),
+            n_samples=32, n_points=16, is_vary_kernel_hyp=True,
+        )
+        assert ds.data.shape == (32, 16, 1)
+
+
+class TestSplitters:
+    def test_random_indcs_bounds(self):
+        getter = GetRandomIndcs(a=0.1, b=0.5)
+        idcs = getter(4, 100)
+        assert 10 <= idcs.shape[1] <= 50
+        assert idcs.max() < 100
+
+    def test_cntxt_trgt_getter_selects(self):
+        X = torch.arange(24).float().view(2, 12, 1) / 24
+        y = X * 2
+        getter = CntxtTrgtGetter(
+            contexts_getter=GetRandomIndcs(a=2, b=5), targets_getter=get_all_indcs
+        )
+        Xc, Yc, Xt, Yt = getter(X, y)
+        assert torch.equal(Xt, X) and torch.equal(Yt, y)
+        assert torch.allclose(Yc, Xc * 2)
+        assert 2 <= Xc.shape[1] <= 5
+
+    def test_grid_getter_coords_normalized(self):
+        X = torch.rand(2, 3, 8, 8)  # B, C, H, W
+        getter = GridCntxtTrgtGetter(
+            context_masker=RandomMasker(a=5, b=5), target_masker=no_masker
+        )
+        Xc, Yc, Xt, Yt = getter(X)
+        assert Xc.shape == (2, 5, 2)
+        assert Xc.min() >= -1 and Xc.max() <= 1
+        assert Xt.shape == (2, 64, 2)
+        assert Yt.shape == (2, 64, 3)
+
+    def test_grid_getter_upscale_factor(self):
+        X = torch.rand(2, 1, 8, 8)
+        getter = GridCntxtTrgtGetter(
+            context_masker=RandomMasker(a=3, b=3), upscale_factor=1.75
+        )
+        Xc, *_ = getter(X)
+        assert Xc.abs().max() <= 1.75 + 1e-6
+
+
+class TestCollate:
+    def test_batch_abi(self):
+        ds = GPDataset(kernel=RBF(0.2), n_samples=16, n_points=16)
+        collate = cntxt_trgt_collate(
+            CntxtTrgtGetter(
+                contexts_getter=GetRandomIndcs(a=2, b=6), targets_getter=get_all_indcs
+            )
+        )
+        loader = torch.utils.data.DataLoader(ds, batch_size=4, collate_fn=collate)
+        inputs, y = next(iter(loader))
+        assert set(inputs) == {"X_cntxt", "Y_cntxt", "X_trgt", "Y_trgt"}
+        assert inputs["X_trgt"].shape == (4, 16, 1)
+        assert torch.equal(inputs["Y_trgt"], y)
+
+    def test_duplicate_batch(self):
+        ds = GPDataset(kernel=RBF(0.2), n_samples=8, n_points=16)
+        collate = cntxt_trgt_collate(
+            CntxtTrgtGetter(
+                contexts_getter=GetRandomIndcs(a=2, b=6), targets_getter=get_all_indcs
+            ),
+            is_duplicate_batch=True,
+        )
+        loader = torch.utils.data.DataLoader(ds, batch_size=4, collate_fn=collate)
+        inputs, y = next(iter(loader))
+        assert y.shape[0] == 8
+        assert torch.equal(inputs["Y_trgt"][:4], inputs["Y_trgt"][4:])
+
+
+def test_synthetic_images():
+    from npf.data.imgs import SyntheticImages, get_dataset, get_train_test_img_dataset
+
+    ds = SyntheticImages(shape=(3, 16, 16), n_samples=8)
+    img, label = ds[0]
+    assert img.shape == (3, 16, 16)
+    assert 0 <= img.min() and img.max() <= 1
+    cls = get_dataset("synthetic32")
+    assert cls.shape == (3, 32, 32)
+    tr, te = get_train_test_img_dataset("synthetic32")
+    assert len(tr) > 0 and len(te) > 0

@@ -1,0 +1,136 @@
+"""Model-layer tests: published parameter counts, forward contract, edge cases."""
+
+import warnings
+
+import pytest
+import torch
+
+from npf import CNPFLoss, ELBOLossLNPF, NLLLossLNPF
+
+from model_zoo import BUILDERS, PUBLISHED_PARAM_COUNTS
+
+warnings.filterwarnings("ignore")
+
+
+@pytest.mark.parametrize("name", sorted(PUBLISHED_PARAM_COUNTS))
+def test_param_counts_match_reference_published(name):
+    """Parameter counts equal the counts printed by the reference notebooks
+    (BASELINE.md 'Parameter counts') — a full structural check."""
+    model = BUILDERS[name]()
+    n = sum(p.numel() for p in model.parameters())
+    assert n == PUBLISHED_PARAM_COUNTS[name]
+
+
+def _set_batch(B=3, C=9, T=17, x_dim=1, y_dim=1):
+    g = torch.Generator().manual_seed(0)
+    Xc = torch.rand(B, C, x_dim, generator=g) * 2 - 1
+    Yc = torch.randn(B, C, y_dim, generator=g)
+    Xt = torch.rand(B, T, x_dim, generator=g) * 2 - 1
+    Yt = torch.randn(B, T, y_dim, generator=g)
+    return Xc, Yc, Xt, Yt
+
+
+@pytest.mark.parametrize(
+    "name,z",
+    [("cnp_1d", 1), ("lnp_1d", 1), ("attncnp_1d", 1), ("attnlnp_1d", 1),
+     ("convcnp_1d", 1), ("convlnp_1d", 16)],
+)
+def test_forward_contract_1d(name, z):
+    model = BUILDERS[name]()
+    model.train()
+    Xc, Yc, Xt, Yt = _set_batch()
+    p_y, z_samples, q_zCc, q_zCct = model(Xc, Yc, Xt, Yt)
+    assert tuple(p_y.batch_shape) == (z, 3, 17)
+    assert tuple(p_y.event_shape) == (1,)
+    if name in ("cnp_1d", "attncnp_1d", "convcnp_1d"):
+        assert z_samples is None and q_zCc is None and q_zCct is None
+    else:
+        assert z_samples is not None and q_zCc is not None
+
+
+@pytest.mark.parametrize("name", ["gridconvcnp_2d", "gridconvlnp_2d"])
+def test_forward_contract_grid(name):
+    model = BUILDERS[name](y_dim=3)
+    model.train()
+    g = torch.Generator().manual_seed(0)
+    B, H, W = 2, 16, 16
+    mask = torch.rand(B, H, W, 1, generator=g) < 0.3
+    img = torch.rand(B, H, W, 3, generator=g)
+    full = torch.ones(B, H, W, 1).bool()
+    p_y, *_ = model(mask, img, full, img)
+    assert tuple(p_y.batch_shape)[1:] == (B, H, W)
+    assert tuple(p_y.event_shape) == (3,)
+
+
+@pytest.mark.parametrize("name", ["cnp_1d", "attncnp_1d", "convcnp_1d"])
+def test_zero_context(name):
+    """Zero-context tasks return a valid distribution (reference np.py:97-99,
+    attnnp.py:108-125, convnp.py:146-151)."""
+    model = BUILDERS[name]()
+    model.eval()
+    _, _, Xt, _ = _set_batch()
+    p_y, *_ = model(torch.zeros(3, 0, 1), torch.zeros(3, 0, 1), Xt)
+    assert tuple(p_y.batch_shape) == (1, 3, 17)
+    assert torch.isfinite(p_y.base_dist.loc).all()
+
+
+def test_training_input_range_validated():
+    model = BUILDERS["cnp_1d"]()
+    model.train()
+    Xc, Yc, Xt, Yt = _set_batch()
+    with pytest.raises(ValueError):
+        model(Xc * 3, Yc, Xt, Yt)
+    model.eval()
+    model(Xc * 3, Yc, Xt)  # no validation at eval time
+
+
+def test_backward_all_params_get_grads():
+    model = BUILDERS["attnlnp_1d"]()
+    model.train()
+    Xc, Yc, Xt, Yt = _set_batch()
+    out = model(Xc, Yc, Xt, Yt)
+    loss = ELBOLossLNPF()(out, Yt)
+    loss.backward()
+    missing = [n for n, p in model.named_parameters() if p.grad is None]
+    assert missing == []
+
+
+def test_convlnp_nll_loss_backward():
+    model = BUILDERS["convlnp_1d"]()
+    model.train()
+    Xc, Yc, Xt, Yt = _set_batch()
+    out = model(Xc, Yc, Xt, Yt)
+    loss = NLLLossLNPF()(out, Yt)
+    loss.backward()
+    assert torch.isfinite(loss)
+
+
+def test_set_extrapolation_regrids():
+    model = BUILDERS["convcnp_1d"]()
+    n0 = model.n_induced
+    model.set_extrapolation([-1, 3])
+    assert model.n_induced != n0
+    model.eval()
+    Xc, Yc, _, _ = _set_batch()
+    Xt = torch.rand(3, 7, 1) * 4 - 1  # beyond [-1, 1]
+    p_y, *_ = model(Xc, Yc, Xt)
+    assert torch.isfinite(p_y.base_dist.loc).all()
+
+
+def test_heteroskedastic_flag_pools_scale():
+    m = BUILDERS["cnp_1d"]()
+    m.is_heteroskedastic = False
+    m.eval()
+    Xc, Yc, Xt, _ = _set_batch()
+    p_y, *_ = m(Xc, Yc, Xt)
+    s = p_y.base_dist.scale
+    assert torch.allclose(s, s.mean(dim=2, keepdim=True).expand_as(s))
+
+
+def test_state_dict_roundtrip():
+    m1 = BUILDERS["convlnp_1d"]()
+    m2 = BUILDERS["convlnp_1d"]()
+    m2.load_state_dict(m1.state_dict())
+    for (n1, p1), (n2, p2) in zip(m1.named_parameters(), m2.named_parameters()):
+        assert n1 == n2
+        assert torch.equal(p1, p2)

@@ -1,0 +1,179 @@
+"""Ops-layer tests: torch reference correctness + autograd, and (on GPU)
+HIP-kernel parity against the fp32 composed-torch oracle."""
+
+import math
+
+import pytest
+import torch
+
+from npf.ops import functional as F_ops
+
+
+def _attn_oracle(keys, queries, values, scale):
+    logits = torch.einsum("bkd,bqd->bqk", keys, queries) * scale
+    return torch.bmm(logits.softmax(-1), values)
+
+
+def _setconv_oracle(keys, queries, values, sigma):
+    diff = keys.unsqueeze(1) - queries.unsqueeze(2)
+    dist = torch.norm(diff, p=2, dim=-1, keepdim=True)
+    inp = -((dist / sigma) ** 2)
+    w = torch.softmax(inp, dim=-2)
+    density = torch.exp(inp).sum(dim=-2)
+    out = (w * values.unsqueeze(1)).sum(dim=2)
+    return torch.cat([out, density], dim=-1)
+
+
+class TestAttention:
+    def test_matches_oracle(self):
+        g = torch.Generator().manual_seed(0)
+        k = torch.randn(4, 9, 16, generator=g)
+        q = torch.randn(4, 33, 16, generator=g)
+        v = torch.randn(4, 9, 16, generator=g)
+        scale = 1 / math.sqrt(16)
+        out = F_ops.attention_qkv(k, q, v, scale)
+        assert torch.allclose(out, _attn_oracle(k, q, v, scale), atol=1e-6)
+
+    def test_autograd(self):
+        g = torch.Generator().manual_seed(1)
+        k = torch.randn(2, 5, 8, generator=g, dtype=torch.float64, requires_grad=True)
+        q = torch.randn(2, 7, 8, generator=g, dtype=torch.float64, requires_grad=True)
+        v = torch.randn(2, 5, 8, generator=g, dtype=torch.float64, requires_grad=True)
+        assert torch.autograd.gradcheck(
+            lambda k, q, v: F_ops.attention_qkv(k, q, v, 0.35), (k, q, v)
+        )
+
+
+class TestSetConv:
+    def test_matches_oracle(self):
+        g = torch.Generator().manual_seed(0)
+        k = torch.rand(3, 11, 1, generator=g) * 2 - 1
+        q = torch.rand(3, 19, 1, generator=g) * 2 - 1
+        v = torch.randn(3, 11, 4, generator=g)
+        sigma = torch.tensor(0.13)
+        out = F_ops.setconv_gaussian(k, q, v, sigma)
+        assert torch.allclose(out, _setconv_oracle(k, q, v, sigma), atol=1e-6)
+
+    def test_autograd_matches_composed(self):
+        g = torch.Generator().manual_seed(2)
+        k = torch.rand(2, 6, 1, generator=g, requires_grad=True)
+        q = torch.rand(2, 9, 1, generator=g, requires_grad=True)
+        v = torch.randn(2, 6, 3, generator=g, requires_grad=True)
+        sigma = torch.tensor(0.2, requires_grad=True)
+        out = F_ops.setconv_gaussian(k, q, v, sigma)
+        loss = (out * torch.arange(out.numel()).float().view_as(out)).sum()
+        gk, gq, gv, gs = torch.autograd.grad(loss, (k, q, v, sigma))
+
+        k2 = k.detach().requires_grad_(True)
+        q2 = q.detach().requires_grad_(True)
+        v2 = v.detach().requires_grad_(True)
+        s2 = sigma.detach().requires_grad_(True)
+        out2 = _setconv_oracle(k2, q2, v2, s2)
+        loss2 = (out2 * torch.arange(out2.numel()).float().view_as(out2)).sum()
+        gk2, gq2, gv2, gs2 = torch.autograd.grad(loss2, (k2, q2, v2, s2))
+        for a, b in [(gk, gk2), (gq, gq2), (gv, gv2), (gs, gs2)]:
+            assert torch.allclose(a, b, atol=1e-5), (a - b).abs().max()
+
+
+class TestGaussianLL:
+    def test_matches_distribution_api(self):
+        g = torch.Generator().manual_seed(0)
+        loc = torch.randn(2, 3, 7, 2, generator=g)
+        scale = torch.rand(2, 3, 7, 2, generator=g) + 0.1
+        y = torch.randn(3, 7, 2, generator=g)
+        dist = torch.distributions.Independent(
+            torch.distributions.Normal(loc, scale), 1
+        )
+        expected = dist.log_prob(y).reshape(2, 3, -1).sum(-1)
+        got = F_ops.gaussian_nll_sum(loc, scale, y)
+        assert torch.allclose(got, expected, atol=1e-5)
+
+    def test_autograd(self):
+        g = torch.Generator().manual_seed(1)
+        loc = torch.randn(2, 2, 5, 1, generator=g, dtype=torch.float64,
+                          requires_grad=True)
+        scale = (torch.rand(2, 2, 5, 1, generator=g, dtype=torch.float64) + 0.1
+                 ).requires_grad_(True)
+        y = torch.randn(2, 5, 1, generator=g, dtype=torch.float64)
+        assert torch.autograd.gradcheck(
+            lambda l, s: F_ops.gaussian_nll_sum(l, s, y), (loc, scale)
+        )
+
+
+# --------------------------------------------------------------------------- #
+# GPU: HIP kernels vs fp32 torch oracle
+# --------------------------------------------------------------------------- #
+
+
+@pytest.mark.gpu
+class TestHIPKernels:
+    def setup_method(self):
+        from npf.ops import has_extension
+
+        if not has_extension():
+            pytest.fail("HIP extension not built/loadable on a GPU box")
+
+    @pytest.mark.parametrize("B,K,Q,D", [(8, 13, 128, 16), (32, 50, 128, 16),
+                                         (256, 300, 1024, 16)])
+    def test_attention_fwd_bwd(self, B, K, Q, D):
+        g = torch.Generator(device="cuda").manual_seed(0)
+        k = torch.randn(B, K, D, device="cuda", generator=g, requires_grad=True)
+        q = torch.randn(B, Q, D, device="cuda", generator=g, requires_grad=True)
+        v = torch.randn(B, K, D, device="cuda", generator=g, requires_grad=True)
+        scale = 1 / math.sqrt(D)
+        out = F_ops.attention_qkv(k, q, v, scale)
+        k0 = k.detach().cpu().requires_grad_(True)
+        q0 = q.detach().cpu().requires_grad_(True)
+        v0 = v.detach().cpu().requires_grad_(True)
+        ref = _attn_oracle(k0, q0, v0, scale)
+        assert torch.allclose(out.cpu(), ref, atol=2e-4), (
+            (out.cpu() - ref).abs().max()
+        )
+        dout = torch.randn_like(out)
+        out.backward(dout)
+        ref.backward(dout.cpu())
+        for a, b in [(k.grad, k0.grad), (q.grad, q0.grad), (v.grad, v0.grad)]:
+            assert torch.allclose(a.cpu(), b, atol=2e-3), (a.cpu() - b).abs().max()
+
+    @pytest.mark.parametrize("B,K,Q,C", [(4, 17, 33, 1), (32, 50, 192, 1),
+                                         (64, 192, 128, 128)])
+    def test_setconv_fwd_bwd(self, B, K, Q, C):
+        g = torch.Generator(device="cuda").manual_seed(0)
+        k = (torch.rand(B, K, 1, device="cuda", generator=g) * 2 - 1).requires_grad_(True)
+        q = (torch.rand(B, Q, 1, device="cuda", generator=g) * 2 - 1).requires_grad_(True)
+        v = torch.randn(B, K, C, device="cuda", generator=g, requires_grad=True)
+        sigma = torch.tensor(0.1, device="cuda", requires_grad=True)
+        out = F_ops.setconv_gaussian(k, q, v, sigma)
+        k0 = k.detach().cpu().requires_grad_(True)
+        q0 = q.detach().cpu().requires_grad_(True)
+        v0 = v.detach().cpu().requires_grad_(True)
+        s0 = sigma.detach().cpu().requires_grad_(True)
+        ref = _setconv_oracle(k0, q0, v0, s0)
+        assert torch.allclose(out.cpu(), ref, atol=2e-4), (out.cpu() - ref).abs().max()
+        dout = torch.randn_like(out)
+        out.backward(dout)
+        ref.backward(dout.cpu())
+        for a, b in [(k.grad, k0.grad), (q.grad, q0.grad), (v.grad, v0.grad),
+                     (sigma.grad, s0.grad)]:
+            assert torch.allclose(a.cpu(), b, atol=3e-3), (a.cpu() - b).abs().max()
+
+    @pytest.mark.parametrize("Z,B,T,Y", [(1, 4, 128, 1), (16, 32, 128, 1),
+                                         (8, 16, 1024, 3)])
+    def test_gaussian_ll_fwd_bwd(self, Z, B, T, Y):
+        g = torch.Generator(device="cuda").manual_seed(0)
+        loc = torch.randn(Z, B, T, Y, device="cuda", generator=g, requires_grad=True)
+        scale = (torch.rand(Z, B, T, Y, device="cuda", generator=g) + 0.05
+                 ).requires_grad_(True)
+        y = torch.randn(B, T, Y, device="cuda", generator=g)
+        out = F_ops.gaussian_nll_sum(loc, scale, y)
+        l0 = loc.detach().cpu().requires_grad_(True)
+        s0 = scale.detach().cpu().requires_grad_(True)
+        y0 = y.cpu()
+        dist = torch.distributions.Independent(torch.distributions.Normal(l0, s0), 1)
+        ref = dist.log_prob(y0).reshape(Z, B, -1).sum(-1)
+        assert torch.allclose(out.cpu(), ref, atol=1e-3), (out.cpu() - ref).abs().max()
+        dout = torch.randn_like(out)
+        out.backward(dout)
+        ref.backward(dout.cpu())
+        assert torch.allclose(loc.grad.cpu(), l0.grad, atol=1e-3)
+        assert torch.allclose(scale.grad.cpu(), s0.grad, atol=1e-3)

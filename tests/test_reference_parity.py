@@ -1,0 +1,169 @@
+"""Golden parity vs the reference's shipped pretrained checkpoints.
+
+These tests are the strongest structural checks we have: they load the
+reference repo's `results/pretrained/.../params.pt` (data files, read-only)
+into OUR modules and verify key-space identity; a subprocess runs the
+REFERENCE implementation itself on fixed inputs and we compare outputs.
+Skipped when /root/reference is absent.
+"""
+
+import os
+import subprocess
+import sys
+import warnings
+
+import pytest
+import torch
+
+from model_zoo import BUILDERS
+
+warnings.filterwarnings("ignore")
+
+REF = "/root/reference"
+PRETRAINED = os.path.join(REF, "results", "pretrained", "RBF_Kernel")
+
+needs_ref = pytest.mark.skipif(
+    not os.path.isdir(PRETRAINED), reason="reference pretrained results unavailable"
+)
+
+REF_MODEL_DIRS = {
+    "cnp_1d": "CNP",
+    "lnp_1d": "LNP",
+    "attncnp_1d": "AttnCNP",
+    "attnlnp_1d": "AttnLNP",
+    "convcnp_1d": "ConvCNP",
+    "convlnp_1d": "ConvLNP",
+}
+
+
+@needs_ref
+@pytest.mark.parametrize("name", sorted(REF_MODEL_DIRS))
+def test_checkpoint_key_space_identical(name):
+    sd_ref = torch.load(
+        os.path.join(PRETRAINED, REF_MODEL_DIRS[name], "run_0", "params.pt"),
+        map_location="cpu",
+    )
+    model = BUILDERS[name]()
+    sd_mine = model.state_dict()
+    assert set(sd_ref) == set(sd_mine)
+    for k in sd_ref:
+        assert sd_ref[k].shape == sd_mine[k].shape, k
+    model.load_state_dict(sd_ref)  # must not raise
+
+
+_REF_RUNNER = r"""
+import sys, warnings, torch
+warnings.filterwarnings("ignore")
+sys.path.insert(0, "/root/reference")
+from functools import partial
+from npf import CNP, AttnCNP, ConvCNP
+from npf.architectures import MLP, merge_flat_input, CNN, ResConvBlock, SetConv, discard_ith_arg
+
+inp = torch.load(sys.argv[1], weights_only=False)
+Xc, Yc, Xt = inp["Xc"], inp["Yc"], inp["Xt"]
+R = 128
+models = {
+ "cnp_1d": CNP(x_dim=1, y_dim=1,
+   XYEncoder=merge_flat_input(partial(MLP, n_hidden_layers=2, hidden_size=R*2), is_sum_merge=True),
+   XEncoder=partial(MLP, n_hidden_layers=1, hidden_size=R),
+   Decoder=merge_flat_input(partial(MLP, n_hidden_layers=4, hidden_size=R), is_sum_merge=True), r_dim=R),
+ "attncnp_1d": AttnCNP(x_dim=1, y_dim=1,
+   XYEncoder=merge_flat_input(partial(MLP, n_hidden_layers=2, hidden_size=R), is_sum_merge=True),
+   is_self_attn=False, r_dim=R, attention="transformer",
+   XEncoder=partial(MLP, n_hidden_layers=1, hidden_size=R),
+   Decoder=merge_flat_input(partial(MLP, n_hidden_layers=4, hidden_size=R), is_sum_merge=True)),
+ "convcnp_1d": ConvCNP(x_dim=1, y_dim=1, Interpolator=SetConv,
+   CNN=partial(CNN, Conv=torch.nn.Conv1d, Normalization=torch.nn.BatchNorm1d, n_blocks=5,
+               kernel_size=19, ConvBlock=ResConvBlock, is_chan_last=True, n_conv_layers=2),
+   density_induced=64, r_dim=R,
+   Decoder=discard_ith_arg(partial(MLP, n_hidden_layers=4, hidden_size=R), i=0)),
+}
+ckpt = {"cnp_1d": "CNP", "attncnp_1d": "AttnCNP", "convcnp_1d": "ConvCNP"}
+out = {}
+for name, m in models.items():
+    sd = torch.load(f"/root/reference/results/pretrained/RBF_Kernel/{ckpt[name]}/run_0/params.pt",
+                    map_location="cpu")
+    m.load_state_dict(sd); m.eval()
+    with torch.no_grad():
+        p, *_ = m(Xc, Yc, Xt)
+    out[name] = (p.base_dist.loc, p.base_dist.scale)
+torch.save(out, sys.argv[2])
+"""
+
+
+@needs_ref
+def test_forward_matches_reference_implementation(tmp_path):
+    """Same pretrained weights + same inputs through the reference code and
+    ours: CNP/AttnCNP bitwise, ConvCNP to fp32 roundoff."""
+    g = torch.Generator().manual_seed(42)
+    Xc = torch.rand(3, 11, 1, generator=g) * 2 - 1
+    Yc = torch.randn(3, 11, 1, generator=g)
+    Xt = torch.rand(3, 37, 1, generator=g) * 2 - 1
+    inp_path = str(tmp_path / "inp.pt")
+    out_path = str(tmp_path / "out.pt")
+    torch.save({"Xc": Xc, "Yc": Yc, "Xt": Xt}, inp_path)
+
+    env = dict(os.environ, PYTHONPATH=REF)
+    subprocess.run(
+        [sys.executable, "-c", _REF_RUNNER, inp_path, out_path],
+        check=True, cwd="/tmp", env=env, capture_output=True,
+    )
+    golden = torch.load(out_path, weights_only=False)
+
+    tol = {"cnp_1d": 0.0, "attncnp_1d": 0.0, "convcnp_1d": 1e-4}
+    for name, builder in (
+        ("cnp_1d", BUILDERS["cnp_1d"]),
+        ("attncnp_1d", BUILDERS["attncnp_1d"]),
+        ("convcnp_1d", BUILDERS["convcnp_1d"]),
+    ):
+        m = builder()
+        sd = torch.load(
+            os.path.join(PRETRAINED, REF_MODEL_DIRS[name], "run_0", "params.pt"),
+            map_location="cpu",
+        )
+        m.load_state_dict(sd)
+        m.eval()
+        with torch.no_grad():
+            p, *_ = m(Xc, Yc, Xt)
+        rl, rs = golden[name]
+        assert (p.base_dist.loc - rl).abs().max() <= tol[name]
+        assert (p.base_dist.scale - rs).abs().max() <= tol[name]
+
+
+@needs_ref
+def test_eval_loglike_matches_published_mean():
+    """Re-evaluating a shipped checkpoint on tasks drawn from the matching GP
+    reproduces the published mean test log-likelihood (AttnCNP
+    on RBF = 149.16) within sampling error of a 512-task draw."""
+    import numpy as np
+
+    from npf import CNPFLoss
+    from npf.data import GPDataset, cntxt_trgt_collate
+    from npf.data.kernels import RBF
+    from npf.train import NPFTrainer, eval_loglike
+    from npf.utils.datasplit import CntxtTrgtGetter, GetRandomIndcs, get_all_indcs
+
+    model = BUILDERS["attncnp_1d"]()
+    sd = torch.load(
+        os.path.join(PRETRAINED, "AttnCNP", "run_0", "params.pt"), map_location="cpu"
+    )
+    model.load_state_dict(sd)
+
+    ds = GPDataset(kernel=RBF(length_scale=0.2), n_samples=2048, n_points=128)
+    collate = cntxt_trgt_collate(
+        CntxtTrgtGetter(
+            contexts_getter=GetRandomIndcs(a=0.0, b=50), targets_getter=get_all_indcs
+        )
+    )
+    # batch 16 => 128 independent context-count draws: the per-task LL swings
+    # ~-160 (1 context) to ~+360 (50 contexts) nats, so the estimator's noise
+    # is dominated by the count draws shared within each batch
+    trainer = NPFTrainer(
+        model, CNPFLoss(), collate_fn=collate, device="cpu", batch_size=16,
+        valid_batch_size=16,
+    )
+    ll = eval_loglike(trainer, ds, seed=123)
+    mean = float(np.mean(ll))
+    # published 149.16 on 10k tasks; ~128 count draws put the mean inside
+    # +-60 nats (3 sigma) unless the model/data pipeline is wrong
+    assert 89 < mean < 209, mean
