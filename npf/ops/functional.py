@@ -78,10 +78,21 @@ def attention_qkv(keys, queries, values, scale=None):
     """softmax(scale * Q K^T) V, softmax over keys.
 
     keys: [N, K, D]; queries: [N, Q, D]; values: [N, K, Dv] -> [N, Q, Dv].
+
+    The fused HIP kernel covers the NP regime (per-head D, Dv <= 32 — every
+    shipped config uses 16); larger head dims use the composed rocBLAS path.
     """
     if scale is None:
         scale = 1.0 / math.sqrt(queries.size(-1))
-    if queries.is_cuda:
+    fusable = (
+        queries.is_cuda
+        and queries.size(-1) <= 32
+        and values.size(-1) <= 32
+        and keys.size(1) > 0
+        and queries.dtype in (torch.float32, torch.bfloat16)
+        and keys.dtype == queries.dtype == values.dtype
+    )
+    if fusable:
         return _AttentionFn.apply(
             keys.contiguous(), queries.contiguous(), values.contiguous(), scale
         )
@@ -148,18 +159,32 @@ class _SetConvFn(torch.autograd.Function):
         )
 
 
+_SETCONV_MAX_K = 4096  # LDS stripe cap of the fused kernel (csrc setconv.hip)
+
+
 def setconv_gaussian(keys, queries, values, sigma):
     """Fused Gaussian SetConv: returns [B, Q, C+1] (weighted values ++ density).
 
     `sigma` is the post-softplus length-scale (a 0-d/1-elem tensor so the
-    learned parameter receives gradient).
+    learned parameter receives gradient).  Positions stay fp32 (length scales
+    are ~4e-3: bf16 positions would destroy the RBF) and bf16 values are
+    upcast before the kernel.
     """
     sigma = sigma.reshape(())
-    if queries.is_cuda:
+    B, K, _ = keys.shape
+    if K == 0:
+        # empty context: zero weighted sum AND zero density (matches the
+        # reference's empty-softmax semantics, setcnn.py:253-266)
+        Q, C = queries.size(1), values.size(2)
+        return queries.new_zeros(B, Q, C + 1) * sigma  # keep sigma in graph
+    if queries.is_cuda and keys.size(2) == 1 and K <= _SETCONV_MAX_K:
         return _SetConvFn.apply(
-            keys.contiguous(), queries.contiguous(), values.contiguous(), sigma
+            keys.float().contiguous(),
+            queries.float().contiguous(),
+            values.float().contiguous(),
+            sigma.float(),
         )
-    return _setconv_ref(keys, queries, values, sigma)
+    return _setconv_ref(keys.float(), queries.float(), values.float(), sigma)
 
 
 # --------------------------------------------------------------------------- #
@@ -216,6 +241,7 @@ def gaussian_nll_sum(loc, scale, y):
     if y.dim() == loc.dim() - 1:
         y = y.unsqueeze(0)
     y = y.expand_as(loc)
+    loc, scale, y = loc.float(), scale.float(), y.float()
     if loc.is_cuda:
         return _GaussLLFn.apply(loc.contiguous(), scale.contiguous(), y.contiguous())
     return _nll_ref(loc, scale, y)
