@@ -241,7 +241,9 @@ def gaussian_nll_sum(loc, scale, y):
     if y.dim() == loc.dim() - 1:
         y = y.unsqueeze(0)
     y = y.expand_as(loc)
-    loc, scale, y = loc.float(), scale.float(), y.float()
-    if loc.is_cuda:
+    if loc.dtype not in (torch.float32, torch.float64):
+        loc, scale = loc.float(), scale.float()  # bf16/half -> fp32 math
+    y = y.to(loc.dtype)
+    if loc.is_cuda and loc.dtype == torch.float32:
         return _GaussLLFn.apply(loc.contiguous(), scale.contiguous(), y.contiguous())
     return _nll_ref(loc, scale, y)
