@@ -151,13 +151,21 @@ def main():
     # ---- optional hipGraph capture of the whole train step ----
     graph = None
     if use_cuda and not args.no_graph:
+        import gc
+
+        # drop every reference to the warmup autograd graph: a live
+        # AccumulateGrad from a pre-capture iteration breaks capture
+        loss = float(loss.detach())
+        gc.collect()
         try:
             s = torch.cuda.Stream()
             s.wait_stream(torch.cuda.current_stream())
             with torch.cuda.stream(s):
                 for i in range(3):
                     load(i)
-                    train_step()
+                    wl = train_step()
+            del wl
+            gc.collect()
             torch.cuda.current_stream().wait_stream(s)
             torch.cuda.synchronize()
             graph = torch.cuda.CUDAGraph()
@@ -213,7 +221,10 @@ def main():
 
     global_batch = args.batch * n_gpus
     tasks_per_sec = global_batch * args.steps / elapsed
-    final_loss = float((static_loss if graph is not None else loss).detach())
+    if graph is not None:
+        final_loss = float(static_loss.detach())
+    else:
+        final_loss = float(loss.detach()) if torch.is_tensor(loss) else float(loss)
 
     if rank == 0:
         result = {

@@ -160,7 +160,7 @@ __global__ void __launch_bounds__(ATTN_BLOCK) npf_attn_bwd_dq(
     T* gp = dq + ((size_t)n * Q + qi) * D;
     #pragma unroll
     for (int d = 0; d < DMAX; ++d)
-      if (d < D) stf(gp + d, acc[d]);  // note: qreg already carried `scale`
+      if (d < D) stf(gp + d, acc[d] * scale);  // dlogits/dq = scale * k
   }
 }
 
