@@ -108,7 +108,9 @@ def main():
     ddp = dist_utils.FlatDDP(model) if (world > 1 or use_cuda) else None
     if use_cuda:
         try:  # fused Adam: one kernel instead of ~500 per-param launches
-            opt = torch.optim.Adam(model.parameters(), lr=LR, fused=True)
+            opt = torch.optim.Adam(
+                model.parameters(), lr=LR, fused=True, capturable=True
+            )
         except Exception:
             opt = torch.optim.Adam(
                 model.parameters(), lr=LR, capturable=True, foreach=True

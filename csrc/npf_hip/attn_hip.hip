@@ -366,7 +366,7 @@ __global__ void __launch_bounds__(ATTN_BLOCK) npf_attn_bwd_dkv(
     T* vp = dv + ((size_t)n * K + ki) * Dv;
     #pragma unroll
     for (int d = 0; d < DMAX; ++d) {
-      if (d < D) stf(kp + d, dk0[d] + dk1[d]);
+      if (d < D) stf(kp + d, (dk0[d] + dk1[d]) * scale);  // dlogits/dk = scale*q
       if (d < Dv) stf(vp + d, dv0[d] + dv1[d]);
     }
   }
