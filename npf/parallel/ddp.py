@@ -100,11 +100,13 @@ class FlatDDP:
     construction so replicas start identical.
     """
 
-    def __init__(self, module, grad_dtype=torch.float32):
+    def __init__(self, module, grad_dtype=None):
         self.module = module
         self.params = [p for p in module.parameters() if p.requires_grad]
         numel = sum(p.numel() for p in self.params)
         device = self.params[0].device if self.params else torch.device("cpu")
+        if grad_dtype is None:
+            grad_dtype = self.params[0].dtype if self.params else torch.float32
         self.flat_grads = torch.zeros(numel, dtype=grad_dtype, device=device)
         # carve the buffer into per-param gradient views
         offset = 0
