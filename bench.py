@@ -166,7 +166,8 @@ def main():
     def train_step():
         if use_bf16:
             # one multi-tensor cast: fp32 master -> bf16 compute weights
-            torch._foreach_copy_(compute_params, master_params)
+            with torch.no_grad():
+                torch._foreach_copy_(compute_params, master_params)
         if ddp is not None:
             ddp.zero_grad_()
         else:
