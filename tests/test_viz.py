@@ -168,7 +168,7 @@ def test_marginal_log_like_and_sarle():
     samples = torch.zeros(1, 100, 1, 1)
     ml = marginal_log_like(pred, samples)
     # z-marginal of N(0,1) at 0 with 4 identical comps = pdf(0) = 1/sqrt(2pi)
-    assert torch.allclose(ml, torch.tensor(1 / np.sqrt(2 * np.pi)), atol=1e-4)
+    assert torch.allclose(ml, torch.full_like(ml, 1 / float(np.sqrt(2 * np.pi))), atol=1e-4)
 
     unimodal = np.random.RandomState(0).randn(500, 8)
     s = sarle(unimodal)
