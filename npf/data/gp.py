@@ -105,8 +105,24 @@ class GPDataset(Dataset):
 
         self._idx_precompute = 0
         self._idx_chunk = 0
+        self._generator = generator
 
         self.precompute_chunk_()
+
+    @property
+    def generator(self):
+        """sklearn GaussianProcessRegressor with this dataset's kernel — the
+        oracle model for posterior-overlay plots (reference
+        gaussian_process.py exposes the sampling GPR the same way)."""
+        if self._generator is None:
+            from sklearn.gaussian_process import GaussianProcessRegressor
+
+            from .kernels import to_sklearn
+
+            self._generator = GaussianProcessRegressor(
+                kernel=to_sklearn(self.kernel), alpha=1e-5, optimizer=None
+            )
+        return self._generator
 
     # ------------------------------------------------------------------ #
     # Dataset protocol

@@ -159,6 +159,29 @@ class ProductKernel(_Composite):
         return self.k1(X) * self.k2(X)
 
 
+def to_sklearn(kernel):
+    """Convert a torch-native Kernel back to the sklearn equivalent (used for
+    the oracle-GP overlay in 1D posterior plots)."""
+    import sklearn.gaussian_process.kernels as SK
+
+    if isinstance(kernel, SumKernel):
+        return to_sklearn(kernel.k1) + to_sklearn(kernel.k2)
+    if isinstance(kernel, ProductKernel):
+        return to_sklearn(kernel.k1) * to_sklearn(kernel.k2)
+    if isinstance(kernel, Matern):
+        return SK.Matern(length_scale=float(kernel.length_scale), nu=kernel.nu)
+    if isinstance(kernel, RBF):
+        return SK.RBF(length_scale=float(kernel.length_scale))
+    if isinstance(kernel, ExpSineSquared):
+        return SK.ExpSineSquared(
+            length_scale=float(kernel.length_scale),
+            periodicity=float(kernel.periodicity),
+        )
+    if isinstance(kernel, WhiteKernel):
+        return SK.WhiteKernel(noise_level=float(kernel.noise_level))
+    raise ValueError(f"cannot convert {kernel!r} to sklearn")
+
+
 def _bounds_of(sk, name):
     b = getattr(sk, f"{name}_bounds", "fixed")
     if isinstance(b, str):  # "fixed"
