@@ -1,19 +1,28 @@
 #!/usr/bin/env python3
-"""Flagship training benchmark: AttnCNP on 1D RBF-GP regression.
+"""Flagship training benchmarks for the BASELINE.json configs.
 
-Measures whole-job training throughput (tasks/sec; a task = one GP function's
-context->target episode) for the BASELINE.json headline config: AttnCNP,
-r_dim=128, transformer cross-attention, 128 target points, batch 32 per GPU,
-bf16 compute, Adam — the reference's 1D training configuration
-(BASELINE.md 'Training configuration') on synthetic RBF-GP tasks with
-random-init weights (no network egress for datasets).
+Default (driver contract): AttnCNP on 1D RBF-GP regression — whole-job
+training throughput in tasks/sec (a task = one GP function's context->target
+episode), batch 32/GPU, bf16 autocast, Adam, the reference's 1D training
+configuration (BASELINE.md 'Training configuration') on synthetic RBF-GP
+tasks with random-init weights (no network egress for datasets).
+
+`--model` selects the other headline configs (all synthetic-data,
+random-init):
+  attncnp       AttnCNP-1D, RBF GP             (BASELINE config #2, default)
+  convcnp       ConvCNP-1D, periodic GP        (BASELINE config #3)
+  attnlnp2d     AttnLNP, CelebA32-shape tasks  (BASELINE config #4)
+  gridconvlnp2d GridConvLNP, CelebA64-shape    (BASELINE config #5)
 
 MI355X-first execution:
-- fused HIP kernels for cross-attention (fwd+bwd) and the Gaussian NLL
-  reduction; hipBLASLt bf16 GEMMs for the MLP/projection stack;
+- fused HIP kernels for cross-attention (fwd+bwd), SetConv and the Gaussian
+  NLL reduction; hipBLASLt bf16 GEMMs for the MLP/projection stack;
+- bf16 autocast mixed precision (fp32 master weights + fp32 softmax/
+  reductions/loss): measured to match fp32 convergence where a pure-bf16
+  weight replica oscillates and diverges;
 - the whole train step (forward+loss+backward+Adam [+RCCL all-reduce]) is
-  captured in a hipGraph and replayed — the model is ~250 k params, so the
-  eager step is launch-bound and graph replay is the first-order lever;
+  captured in a hipGraph and replayed — these models are <1 M params, so
+  the eager step is launch-bound and graph replay is the first-order lever;
 - data parallel: one process per GPU, flat-buffer all-reduce over RCCL/xGMI
   (npf.parallel.FlatDDP), rank-offset task sampling (weak scaling).
 
@@ -23,7 +32,6 @@ prints ONE JSON line.
 
 import argparse
 import json
-import math
 import os
 import sys
 import time
@@ -33,12 +41,8 @@ import torch
 sys.path.insert(0, os.path.abspath(os.path.dirname(__file__)))
 sys.path.insert(0, os.path.join(os.path.abspath(os.path.dirname(__file__)), "tests"))
 
-N_POINTS = 128      # target points per task (reference 1D config)
-N_CNTXT = 50        # context points: max of the reference's U(0,50) draw
-                    # (fixed for static graph shapes; max = most work)
-BATCH_PER_GPU = 32  # reference 1D batch size
 LR = 1e-3
-POOL_BATCHES = 16   # pre-generated synthetic task pool (cycled)
+POOL_BATCHES = 16  # pre-generated synthetic task pool (cycled)
 
 
 def log(msg):
@@ -46,34 +50,146 @@ def log(msg):
         print(msg, file=sys.stderr, flush=True)
 
 
-def make_task_pool(device, batch, seed):
-    """Pre-draw a pool of synthetic RBF-GP task batches on-device.
+# --------------------------------------------------------------------------- #
+# synthetic task pools (generated on-device, reference task distributions)
+# --------------------------------------------------------------------------- #
 
-    Batched-Cholesky GP prior draws (RBF length_scale=0.2 on [-2,2],
-    rescaled to [-1,1]) — the same task distribution the reference trains on
-    (utils/ntbks_helpers.py:78-99), generated on the GPU.
-    """
-    from npf.data.kernels import RBF
 
+def _gp_pool(device, batch, seed, kernel, n_points=128, n_cntxt=50):
+    """Batched-Cholesky GP prior draws on [-2,2] rescaled to [-1,1] — the
+    reference 1D training distribution (utils/ntbks_helpers.py:78-99)."""
     g = torch.Generator(device="cpu").manual_seed(seed)
-    kernel = RBF(length_scale=0.2)
     pool = []
     for _ in range(POOL_BATCHES):
-        X = torch.empty(batch, N_POINTS, 1).uniform_(-2, 2, generator=g)
+        X = torch.empty(batch, n_points, 1).uniform_(-2, 2, generator=g)
         X, _ = X.sort(dim=1)
         cov = kernel(X.double())
         L = torch.linalg.cholesky(
-            cov + 1e-6 * torch.eye(N_POINTS, dtype=torch.float64)
+            cov + 1e-6 * torch.eye(n_points, dtype=torch.float64)
         )
-        eps = torch.randn(batch, N_POINTS, 1, dtype=torch.float64, generator=g)
+        eps = torch.randn(batch, n_points, 1, dtype=torch.float64, generator=g)
         Y = (L @ eps).float()
-        X = (X / 2.0).float()  # rescale [-2,2] -> [-1,1]
-        perm = torch.stack([torch.randperm(N_POINTS, generator=g) for _ in range(batch)])
-        cidx = perm[:, :N_CNTXT]
+        X = (X / 2.0).float()
+        perm = torch.stack(
+            [torch.randperm(n_points, generator=g) for _ in range(batch)]
+        )
+        cidx = perm[:, :n_cntxt]
         Xc = torch.gather(X, 1, cidx.unsqueeze(-1))
         Yc = torch.gather(Y, 1, cidx.unsqueeze(-1))
         pool.append(tuple(t.to(device) for t in (Xc, Yc, X, Y)))
     return pool
+
+
+def make_task_pool(device, batch, seed):
+    """Default AttnCNP pool (back-compat name used by diagnostics)."""
+    from npf.data.kernels import RBF
+
+    return _gp_pool(device, batch, seed, RBF(length_scale=0.2))
+
+
+def _img_point_pool(device, batch, seed, shape=(3, 32, 32), cntxt_frac=0.3):
+    """Point-set image-completion tasks (AttnLNP celeba32-shape): smooth
+    synthetic images, coordinates in [-1,1]^2, context = random subset."""
+    import torch.nn.functional as F
+
+    g = torch.Generator(device="cpu").manual_seed(seed)
+    c, h, w = shape
+    n_pix = h * w
+    ys, xs = torch.meshgrid(
+        torch.linspace(-1, 1, h), torch.linspace(-1, 1, w), indexing="ij"
+    )
+    coords = torch.stack([ys, xs], dim=-1).view(1, n_pix, 2)
+    n_cntxt = int(cntxt_frac * n_pix)
+    pool = []
+    for _ in range(POOL_BATCHES):
+        low = torch.rand(batch, c, h // 4, w // 4, generator=g)
+        img = F.interpolate(low, size=(h, w), mode="bilinear", align_corners=False)
+        Y = img.permute(0, 2, 3, 1).reshape(batch, n_pix, c)
+        X = coords.expand(batch, n_pix, 2)
+        cidx = torch.stack(
+            [torch.randperm(n_pix, generator=g)[:n_cntxt] for _ in range(batch)]
+        )
+        Xc = torch.gather(X, 1, cidx.unsqueeze(-1).expand(-1, -1, 2))
+        Yc = torch.gather(Y, 1, cidx.unsqueeze(-1).expand(-1, -1, c))
+        pool.append(tuple(t.to(device).contiguous() for t in (Xc, Yc, X, Y)))
+    return pool
+
+
+def _img_grid_pool(device, batch, seed, shape=(3, 64, 64), cntxt_frac=0.1):
+    """On-the-grid image tasks (GridConv models): X is a boolean mask."""
+    import torch.nn.functional as F
+
+    g = torch.Generator(device="cpu").manual_seed(seed)
+    c, h, w = shape
+    pool = []
+    for _ in range(POOL_BATCHES):
+        low = torch.rand(batch, c, h // 4, w // 4, generator=g)
+        img = F.interpolate(low, size=(h, w), mode="bilinear", align_corners=False)
+        Y = img.permute(0, 2, 3, 1).contiguous()  # [B,H,W,C]
+        n_cntxt = int(cntxt_frac * h * w)
+        mask_c = torch.zeros(batch, h * w, 1, dtype=torch.bool)
+        for b in range(batch):
+            mask_c[b, torch.randperm(h * w, generator=g)[:n_cntxt]] = True
+        mask_c = mask_c.view(batch, h, w, 1)
+        mask_t = torch.ones(batch, h, w, 1, dtype=torch.bool)
+        pool.append(tuple(t.to(device) for t in (mask_c, Y, mask_t, Y)))
+    return pool
+
+
+# --------------------------------------------------------------------------- #
+# benchmark configurations (BASELINE.json "configs")
+# --------------------------------------------------------------------------- #
+
+
+def _configs():
+    from npf import CNPFLoss, ELBOLossLNPF, NLLLossLNPF
+    from npf.data.kernels import RBF, ExpSineSquared
+    import model_zoo as zoo
+
+    return {
+        "attncnp": dict(
+            builder=zoo.attncnp_1d,
+            loss=CNPFLoss,
+            batch=32,
+            pool=lambda dev, b, s: _gp_pool(dev, b, s, RBF(length_scale=0.2)),
+            needs_y_trgt=False,
+            desc="AttnCNP-1D (r_dim=128, transformer attention, 252,738 params)",
+            seq_len=128,
+            n_cntxt=50,
+        ),
+        "convcnp": dict(
+            builder=zoo.convcnp_1d,
+            loss=CNPFLoss,
+            batch=32,
+            pool=lambda dev, b, s: _gp_pool(
+                dev, b, s, ExpSineSquared(length_scale=0.5, periodicity=0.5)
+            ),
+            needs_y_trgt=False,
+            desc="ConvCNP-1D (r_dim=128, SetConv + 5-block ResConv CNN k=19, 276,612 params)",
+            seq_len=128,
+            n_cntxt=50,
+        ),
+        "attnlnp2d": dict(
+            builder=zoo.attnlnp_2d,
+            loss=ELBOLossLNPF,
+            batch=32,
+            pool=lambda dev, b, s: _img_point_pool(dev, b, s, shape=(3, 32, 32)),
+            needs_y_trgt=True,  # NPVI: q_zCct from the target set
+            desc="AttnLNP-2D CelebA32-shape (self-attn encoder, NPVI, 468,486 params)",
+            seq_len=32 * 32,
+            n_cntxt=int(0.3 * 32 * 32),
+        ),
+        "gridconvlnp2d": dict(
+            builder=zoo.gridconvlnp_2d,
+            loss=NLLLossLNPF,
+            batch=16,
+            pool=lambda dev, b, s: _img_grid_pool(dev, b, s, shape=(3, 64, 64)),
+            needs_y_trgt=False,  # NPML
+            desc="GridConvLNP-2D CelebA64-shape (4+4-block CNN k=9, NPML z=16, 487,793 params)",
+            seq_len=64 * 64,
+            n_cntxt=int(0.1 * 64 * 64),
+        ),
+    }
 
 
 def main():
@@ -81,21 +197,24 @@ def main():
     p.add_argument("--gpus", type=int, default=1)
     p.add_argument("--steps", type=int, default=200)
     p.add_argument("--warmup", type=int, default=20)
-    p.add_argument("--batch", type=int, default=BATCH_PER_GPU)
+    p.add_argument("--model", default="attncnp",
+                   choices=["attncnp", "convcnp", "attnlnp2d", "gridconvlnp2d"])
+    p.add_argument("--batch", type=int, default=None)
     p.add_argument("--no-graph", action="store_true")
     p.add_argument("--profile-tag", default=None, help="roctx-range tag")
     args = p.parse_args()
 
-    from npf import CNPFLoss
     from npf.parallel import ddp as dist_utils
-    from model_zoo import attncnp_1d
+
+    cfg = _configs()[args.model]
+    batch = args.batch or cfg["batch"]
 
     rank, world, local_rank = dist_utils.init_distributed()
     use_cuda = torch.cuda.is_available()
     device = torch.device(f"cuda:{local_rank}" if use_cuda else "cpu")
     if world > 1:
         assert world == args.gpus or args.gpus == 1, (world, args.gpus)
-    n_gpus = world if world > 1 else (1 if use_cuda else 1)
+    n_gpus = world if world > 1 else 1
 
     if use_cuda:
         # per-shape GEMM autotuning (rocBLAS/hipBLASLt/CK): tunes during
@@ -107,8 +226,8 @@ def main():
             pass
 
     torch.manual_seed(123 + rank)
-    model = attncnp_1d().to(device)
-    crit = CNPFLoss()
+    model = cfg["builder"]().to(device)
+    crit = cfg["loss"]()
     crit.train()
     model.train()
 
@@ -120,51 +239,50 @@ def main():
     # kernels are captured inside the hipGraph so their launch cost is
     # amortized to pure GPU time.)
     use_bf16 = use_cuda
-    master_params = [p for p in model.parameters() if p.requires_grad]
+    master_params = [p_ for p_ in model.parameters() if p_.requires_grad]
 
     # flat fp32 grad buffer: one fill for zero_grad, grads accumulate into
     # views, one RCCL all-reduce at world>1, fused Adam reads the views
     ddp = dist_utils.FlatDDP(model)
     if use_cuda:
         try:  # fused Adam: one kernel instead of ~500 per-param launches
-            opt = torch.optim.Adam(
-                master_params, lr=LR, fused=True, capturable=True
-            )
+            opt = torch.optim.Adam(master_params, lr=LR, fused=True, capturable=True)
         except Exception:
-            opt = torch.optim.Adam(
-                master_params, lr=LR, capturable=True, foreach=True
-            )
+            opt = torch.optim.Adam(master_params, lr=LR, capturable=True, foreach=True)
     else:
         opt = torch.optim.Adam(master_params, lr=LR)
 
-    log(f"[bench] generating task pool on {device} ...")
-    pool = make_task_pool(device, args.batch, seed=1234 + rank)
+    log(f"[bench] generating {args.model} task pool on {device} ...")
+    pool = cfg["pool"](device, batch, 1234 + rank)
 
     # static input buffers (graph-capture friendly)
     sXc, sYc, sXt, sYt = (torch.empty_like(t) for t in pool[0])
+    pass_y = cfg["needs_y_trgt"]
 
     def train_step():
         ddp.zero_grad_()
-        # Y_trgt is ignored by the deterministic path (reference base.py:223)
         if use_bf16:
             with torch.autocast("cuda", dtype=torch.bfloat16):
-                out = model(X_cntxt=sXc, Y_cntxt=sYc, X_trgt=sXt)
+                out = model(
+                    X_cntxt=sXc, Y_cntxt=sYc, X_trgt=sXt,
+                    Y_trgt=sYt if pass_y else None,
+                )
         else:
-            out = model(X_cntxt=sXc, Y_cntxt=sYc, X_trgt=sXt)
-        loss = crit(out, sYt)  # fp32 log-prob + reduction
+            out = model(
+                X_cntxt=sXc, Y_cntxt=sYc, X_trgt=sXt,
+                Y_trgt=sYt if pass_y else None,
+            )
+        loss = crit(out, sYt.float())  # fp32 log-prob + reduction
         loss.backward()
         ddp.reduce_()
         opt.step()
         return loss
 
     def load(i):
-        Xc, Yc, Xt, Yt = pool[i % POOL_BATCHES]
-        sXc.copy_(Xc)
-        sYc.copy_(Yc)
-        sXt.copy_(Xt)
-        sYt.copy_(Yt)
+        for buf, t in zip((sXc, sYc, sXt, sYt), pool[i % POOL_BATCHES]):
+            buf.copy_(t)
 
-    # ---- warmup (also primes cuBLAS/MIOpen algo caches) ----
+    # ---- warmup (also primes BLAS/MIOpen algo caches) ----
     for i in range(max(args.warmup, 3)):
         load(i)
         loss = train_step()
@@ -236,13 +354,13 @@ def main():
 
     # max over ranks
     if world > 1:
-        t = torch.tensor([elapsed], device=device if use_cuda else "cpu")
         import torch.distributed as td
 
+        t = torch.tensor([elapsed], device=device if use_cuda else "cpu")
         td.all_reduce(t, op=td.ReduceOp.MAX)
         elapsed = float(t)
 
-    global_batch = args.batch * n_gpus
+    global_batch = batch * n_gpus
     tasks_per_sec = global_batch * args.steps / elapsed
     if graph is not None:
         final_loss = float(static_loss.detach())
@@ -264,10 +382,10 @@ def main():
             "dtype": "bf16",
             "data": "synthetic",
             "config": {
-                "model": "AttnCNP-1D (r_dim=128, transformer attention, 252,738 params)",
+                "model": cfg["desc"],
                 "global_batch": global_batch,
-                "seq_len": N_POINTS,
-                "n_cntxt": N_CNTXT,
+                "seq_len": cfg["seq_len"],
+                "n_cntxt": cfg["n_cntxt"],
                 "parallelism": f"dp{n_gpus}",
                 "graph": graph is not None,
                 "final_loss": final_loss,
