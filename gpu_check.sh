@@ -3,18 +3,24 @@ set -x
 mkdir -p gpurun_out
 R=$GRAFT_REPO_ROOT
 rm -rf gpurun_out/prof
-timeout 600 python bench.py --steps 600 --warmup 50 > gpurun_out/bench.json 2> gpurun_out/bench.log
-echo "bench600: $?" | tee gpurun_out/summary.txt
-cat gpurun_out/bench.json | tee -a gpurun_out/summary.txt
-timeout 300 python -m pytest tests -m gpu -x -q > gpurun_out/pytest_gpu.log 2>&1
-echo "pytest-gpu: $?" | tee -a gpurun_out/summary.txt
-tail -2 gpurun_out/pytest_gpu.log >> gpurun_out/summary.txt
+for m in attncnp convcnp attnlnp2d gridconvlnp2d; do
+  case $m in
+    attncnp|convcnp) steps=300;;
+    attnlnp2d) steps=200;;
+    gridconvlnp2d) steps=100;;
+  esac
+  timeout 600 python bench.py --model $m --steps $steps --warmup 30 > gpurun_out/bench_$m.json 2> gpurun_out/bench_$m.log
+  echo "bench-$m: $?" | tee -a gpurun_out/summary.txt
+  cat gpurun_out/bench_$m.json >> gpurun_out/summary.txt
+done
 cd /tmp && export TMPDIR=/tmp
-timeout 600 rocprofv3 --kernel-trace --stats -d $R/gpurun_out/prof -o attncnp -- python $R/bench.py --steps 100 --warmup 30 > $R/gpurun_out/bench_prof.json 2> $R/gpurun_out/prof.log
-echo "rocprof: $?" >> $R/gpurun_out/summary.txt
+for m in convcnp attnlnp2d; do
+  timeout 600 rocprofv3 --kernel-trace --stats --output-format csv -d $R/gpurun_out/prof -o $m -- python $R/bench.py --model $m --steps 60 --warmup 20 > $R/gpurun_out/prof_$m.json 2> $R/gpurun_out/prof_$m.log
+  echo "rocprof-$m: $?" >> $R/gpurun_out/summary.txt
+done
 cd $R
-# keep only the small stats/summary files: the merge-back cap is 64 MiB
-find gpurun_out/prof -type f -size +4M -delete
-find gpurun_out/prof -name '*kernel_trace*' -delete
+find gpurun_out/prof -name '*kernel_trace.csv' -delete
+find gpurun_out/prof -type f -size +8M -delete
+ls -la gpurun_out/prof >> gpurun_out/summary.txt
 du -sh gpurun_out >> gpurun_out/summary.txt
-cat gpurun_out/summary.txt
+tail -30 gpurun_out/summary.txt
