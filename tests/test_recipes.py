@@ -1,0 +1,128 @@
+"""Experiment-recipe layer tests (reference utils/ntbks_helpers.py parity):
+dataset registries, renamer, y-dim injection, n_cntxt splitters and the
+multi-model comparison plots."""
+
+import matplotlib
+
+matplotlib.use("Agg")
+
+import matplotlib.pyplot as plt
+import pytest
+import torch
+
+from npf import recipes
+from npf.data import GPDataset
+
+from model_zoo import attncnp_2d, cnp_1d, gridconvcnp_2d
+
+
+def teardown_function(_):
+    plt.close("all")
+
+
+def test_get_datasets_single_gp_small():
+    d, t, v = recipes.get_datasets_single_gp(
+        n_samples=64, n_points=16, save_file=None
+    )
+    assert set(d) == {"RBF_Kernel", "Periodic_Kernel", "Noisy_Matern_Kernel"}
+    for ds in d.values():
+        assert not ds.is_reuse_across_epochs
+    # test split is a fixed 10k draw in the reference; with save_file=None it
+    # still freezes on one draw
+    for ds in t.values():
+        assert ds.is_reuse_across_epochs
+    for ds in v.values():
+        assert len(ds) == 64 // 10
+
+
+def test_get_all_gp_datasets_names():
+    d, t, v = recipes.get_all_gp_datasets(
+        n_samples=32, n_points=8, save_file=None
+    )
+    assert set(d) == {
+        "RBF_Kernel", "Periodic_Kernel", "Noisy_Matern_Kernel",
+        "Variable_Matern_Kernel", "All_Kernels",
+    }
+    assert len(d["All_Kernels"]) == 3 * 32
+
+
+def test_pretty_renamer():
+    r = recipes.PRETTY_RENAMER
+    assert r["celeba32"] == "CelebA32"
+    assert r["AttnCNP"] == "AttnCNP"
+    assert r["ConvLNP_ELBOTrue_LatLBTrue_SigLBTrue"].startswith("ConvLNP NPVI")
+    assert r[3] == 3
+
+
+def test_add_y_dim():
+    models = {"m": cnp_1d}
+
+    class FakeDs:
+        shape = (3, 32, 32)
+
+    out = recipes.add_y_dim(models, {"celeba32": FakeDs()})
+    assert out["celeba32"]["m"].keywords == {"y_dim": 3}
+
+
+def test_get_n_cntxt_1d():
+    getter = recipes.get_n_cntxt(7)
+    X = torch.rand(2, 32, 1) * 2 - 1
+    Y = torch.randn(2, 32, 1)
+    Xc, Yc, Xt, Yt = getter(X, Y)
+    assert Xc.shape == (2, 7, 1)
+    assert Xt.shape == (2, 32, 1)
+
+
+def test_get_n_cntxt_2d():
+    getter = recipes.get_n_cntxt(9, is_1d=False)
+    X = torch.rand(2, 3, 8, 8)
+    Xc, Yc, Xt, Yt = getter(X)
+    assert Yc.shape == (2, 9, 3)
+
+
+@pytest.fixture(scope="module")
+def tiny_gp():
+    from npf.data.kernels import RBF
+
+    return GPDataset(
+        kernel=RBF(0.2), n_samples=8, n_points=32, is_reuse_across_epochs=True
+    )
+
+
+def test_plot_multi_posterior_samples_1d(tiny_gp):
+    fig = recipes.plot_multi_posterior_samples_1d(
+        {"RBF_Kernel/CNP": cnp_1d()}, {"RBF_Kernel": tiny_gp}, n_cntxt=5,
+        is_plot_generator=True,
+    )
+    assert len(fig.axes) == 1
+
+
+def test_plot_multi_prior_samples_1d(tiny_gp):
+    fig = recipes.plot_multi_prior_samples_1d(
+        {"RBF_Kernel/CNP": cnp_1d()}, {"RBF_Kernel": tiny_gp}
+    )
+    assert len(fig.axes) == 1
+
+
+def test_plot_multi_posterior_samples_imgs():
+    from npf.data.imgs import SyntheticImages
+
+    ds = SyntheticImages(shape=(3, 16, 16), n_samples=8)
+    fig = recipes.plot_multi_posterior_samples_imgs(
+        {"synthetic32/GridConvCNP": gridconvcnp_2d(y_dim=3)},
+        {"synthetic32": ds},
+        n_cntxt=0.2,
+        n_plots=2,
+    )
+    assert len(fig.axes) == 1
+
+
+def test_gp_dataset_generator_oracle(tiny_gp):
+    gen = tiny_gp.generator
+    import sklearn.gaussian_process
+
+    assert isinstance(gen, sklearn.gaussian_process.GaussianProcessRegressor)
+    # oracle can fit+predict on a context set
+    gen.fit([[0.0], [0.5]], [[0.1], [0.2]])
+    mean, std = gen.predict([[0.25]], return_std=True)
+    assert mean.shape[0] == 1
