@@ -116,6 +116,7 @@ class NPFTrainer:
         is_ddp=True,
         num_workers=0,
         is_progressbar=False,
+        profile=False,
     ):
         self.rank = dist_utils.get_rank()
         self.world_size = dist_utils.get_world_size()
@@ -152,6 +153,10 @@ class NPFTrainer:
         self.amp_dtype = amp_dtype
         self.num_workers = num_workers
         self.is_progressbar = is_progressbar
+        # rocprofv3-visible roctx ranges around step phases (SURVEY.md §5.1:
+        # run under `rocprofv3 --kernel-trace --stats -- python ...` and the
+        # ranges delimit forward/backward/optimizer per-kernel attribution)
+        self.profile = profile and torch.cuda.is_available()
 
         self.optimizer = optimizer(self.module.parameters(), lr=lr)
         self.scheduler = None
@@ -202,6 +207,14 @@ class NPFTrainer:
             drop_last=training,
         )
 
+    def _range_push(self, name):
+        if self.profile:
+            torch.cuda.nvtx.range_push(name)  # roctx range on ROCm
+
+    def _range_pop(self):
+        if self.profile:
+            torch.cuda.nvtx.range_pop()
+
     def train_step(self, inputs, y):
         """One optimization step; returns the (scalar) loss."""
         self.module.train()
@@ -211,11 +224,17 @@ class NPFTrainer:
         else:
             self.optimizer.zero_grad(set_to_none=True)
 
+        self._range_push("npf/forward")
         with self._autocast():
             outputs = self.module(**inputs)
         loss = self.criterion(outputs, y)  # loss math in fp32
-        loss.backward()
+        self._range_pop()
 
+        self._range_push("npf/backward")
+        loss.backward()
+        self._range_pop()
+
+        self._range_push("npf/optimizer")
         if self.ddp is not None:
             self.ddp.reduce_()
         if self.grad_clip_norm is not None:
@@ -223,6 +242,7 @@ class NPFTrainer:
                 self.module.parameters(), self.grad_clip_norm
             )
         self.optimizer.step()
+        self._range_pop()
         return loss.detach()
 
     def validation_step(self, inputs, y):
