@@ -38,6 +38,15 @@ void npf_gauss_ll_fwd_launch(const float*, const float*, const float*, float*,
 void npf_gauss_ll_bwd_launch(const float*, const float*, const float*,
                              const float*, float*, float*, long long,
                              long long, hipStream_t);
+void npf_cb_stats_launch(const float*, float*, float*, float*, float*, float*,
+                         int, int, int, float, float, hipStream_t);
+void npf_cb_fwd_launch(const float*, const float*, const float*, const float*,
+                       const float*, const float*, const float*, const float*,
+                       float*, int, int, int, int, hipStream_t);
+void npf_cb_bwd_launch(const float*, const float*, const float*, const float*,
+                       const float*, const float*, const float*, float*,
+                       float*, float*, float*, float*, float*, float*, float*,
+                       int, int, int, int, int, hipStream_t);
 }
 
 namespace {
@@ -186,6 +195,85 @@ std::tuple<torch::Tensor, torch::Tensor> gauss_ll_bwd(torch::Tensor loc,
   return {dloc, dscale};
 }
 
+// fused conv block: stats (training BN) -> (mean, rstd, save_var)
+std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> convblock_stats(
+    torch::Tensor x, torch::Tensor running_mean, torch::Tensor running_var,
+    double eps, double momentum) {
+  check_cuda_contig(x, "x");
+  TORCH_CHECK(x.scalar_type() == torch::kFloat32, "convblock is fp32");
+  const int N = x.size(0), C = x.size(1), L = x.size(2);
+  auto mean = torch::empty({C}, x.options());
+  auto rstd = torch::empty({C}, x.options());
+  auto save_var = torch::empty({C}, x.options());
+  auto stream = at::hip::getCurrentHIPStream();
+  float* rm = running_mean.defined() ? running_mean.data_ptr<float>() : nullptr;
+  float* rv = running_var.defined() ? running_var.data_ptr<float>() : nullptr;
+  npf_cb_stats_launch(x.data_ptr<float>(), mean.data_ptr<float>(),
+                      rstd.data_ptr<float>(), save_var.data_ptr<float>(), rm,
+                      rv, N, C, L, (float)eps, (float)momentum, stream);
+  return {mean, rstd, save_var};
+}
+
+torch::Tensor convblock_fwd(torch::Tensor x, torch::Tensor res,
+                            torch::Tensor w, torch::Tensor b,
+                            torch::Tensor gamma, torch::Tensor beta,
+                            torch::Tensor mean, torch::Tensor rstd) {
+  check_cuda_contig(x, "x");
+  check_cuda_contig(w, "w");
+  TORCH_CHECK(x.scalar_type() == torch::kFloat32, "convblock is fp32");
+  const int N = x.size(0), C = x.size(1), L = x.size(2);
+  const int K = w.size(-1);
+  TORCH_CHECK(K <= 31 && K % 2 == 1, "kernel size must be odd and <= 31");
+  auto y = torch::empty_like(x);
+  auto stream = at::hip::getCurrentHIPStream();
+  npf_cb_fwd_launch(
+      x.data_ptr<float>(), res.defined() ? res.data_ptr<float>() : nullptr,
+      w.data_ptr<float>(),
+      b.defined() ? b.data_ptr<float>() : nullptr,
+      gamma.defined() ? gamma.data_ptr<float>() : nullptr,
+      gamma.defined() ? beta.data_ptr<float>() : nullptr,
+      gamma.defined() ? mean.data_ptr<float>() : nullptr,
+      gamma.defined() ? rstd.data_ptr<float>() : nullptr, y.data_ptr<float>(),
+      N, C, L, K, stream);
+  return y;
+}
+
+std::vector<torch::Tensor> convblock_bwd(torch::Tensor x, torch::Tensor w,
+                                         torch::Tensor dy, torch::Tensor gamma,
+                                         torch::Tensor beta, torch::Tensor mean,
+                                         torch::Tensor rstd, bool has_bias,
+                                         bool training) {
+  check_cuda_contig(x, "x");
+  check_cuda_contig(w, "w");
+  check_cuda_contig(dy, "dy");
+  const int N = x.size(0), C = x.size(1), L = x.size(2);
+  const int K = w.size(-1);
+  const bool has_bn = gamma.defined();
+  auto dact = torch::empty_like(x);
+  auto dx = torch::empty_like(x);
+  auto dw = torch::zeros_like(w);
+  auto db = has_bias ? torch::zeros({C}, x.options()) : torch::Tensor();
+  auto opts = x.options();
+  auto sum_dxhat = has_bn ? torch::zeros({C}, opts) : torch::Tensor();
+  auto sum_dxhat_xhat = has_bn ? torch::zeros({C}, opts) : torch::Tensor();
+  auto dgamma = has_bn ? torch::zeros({C}, opts) : torch::Tensor();
+  auto dbeta = has_bn ? torch::zeros({C}, opts) : torch::Tensor();
+  auto stream = at::hip::getCurrentHIPStream();
+  npf_cb_bwd_launch(
+      x.data_ptr<float>(), w.data_ptr<float>(), dy.data_ptr<float>(),
+      has_bn ? gamma.data_ptr<float>() : nullptr,
+      has_bn ? beta.data_ptr<float>() : nullptr,
+      has_bn ? mean.data_ptr<float>() : nullptr,
+      has_bn ? rstd.data_ptr<float>() : nullptr, dact.data_ptr<float>(),
+      dw.data_ptr<float>(), has_bias ? db.data_ptr<float>() : nullptr,
+      has_bn ? sum_dxhat.data_ptr<float>() : nullptr,
+      has_bn ? sum_dxhat_xhat.data_ptr<float>() : nullptr,
+      has_bn ? dgamma.data_ptr<float>() : nullptr,
+      has_bn ? dbeta.data_ptr<float>() : nullptr, dx.data_ptr<float>(), N, C,
+      L, K, training ? 1 : 0, stream);
+  return {dx, dw, db, dgamma, dbeta};
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -195,5 +283,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("setconv_bwd", &setconv_bwd, "fused Gaussian SetConv backward");
   m.def("gauss_ll_fwd", &gauss_ll_fwd, "fused Gaussian log-lik forward");
   m.def("gauss_ll_bwd", &gauss_ll_bwd, "fused Gaussian log-lik backward");
+  m.def("convblock_stats", &convblock_stats,
+        "fused conv block: per-channel batch stats (+running update)");
+  m.def("convblock_fwd", &convblock_fwd,
+        "fused bn+relu+depthwise-conv(+residual) forward, 1D");
+  m.def("convblock_bwd", &convblock_bwd,
+        "fused conv block backward -> (dx, dw, db, dgamma, dbeta)");
   m.attr("_arch") = "gfx950";
 }

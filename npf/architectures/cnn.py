@@ -129,7 +129,41 @@ class ResConvBlock(nn.Module):
     def reset_parameters(self):
         weights_init(self)
 
+    def _is_fusable_1d(self, X):
+        from npf.ops import has_extension
+
+        return (
+            X.is_cuda
+            and X.dim() == 3
+            and isinstance(self.conv2_depthwise, nn.Conv1d)
+            and isinstance(self.activation, nn.ReLU)
+            and isinstance(self.norm2, (nn.BatchNorm1d, nn.Identity))
+            and has_extension()
+        )
+
     def forward(self, X):
+        if self._is_fusable_1d(X):
+            # fused HIP path (csrc/npf_hip/convblock.hip): bn+relu+dwconv
+            # collapse to 2 kernels; the pointwise stays a library GEMM
+            from npf.ops import conv_block_1d
+
+            if self.n_conv_layers == 2:
+                h = conv_block_1d(
+                    X,
+                    self.conv1.depthwise,
+                    bn=self.norm1 if isinstance(self.norm1, nn.BatchNorm1d) else None,
+                )
+                h = self.conv1.pointwise(h)
+            else:
+                h = X
+            out = conv_block_1d(
+                h,
+                self.conv2_depthwise,
+                bn=self.norm2 if isinstance(self.norm2, nn.BatchNorm1d) else None,
+                residual=X,
+            )
+            return self.conv2_pointwise(out)
+
         out = self.conv1(self.activation(self.norm1(X))) if self.n_conv_layers == 2 else X
         out = self.conv2_depthwise(self.activation(self.norm2(out)))
         # residual added BEFORE the pointwise so out_chan may differ

@@ -12,7 +12,7 @@ import torch
 
 from . import _backend
 
-__all__ = ["attention_qkv", "setconv_gaussian", "gaussian_nll_sum"]
+__all__ = ["attention_qkv", "setconv_gaussian", "gaussian_nll_sum", "conv_block_1d"]
 
 
 # --------------------------------------------------------------------------- #
@@ -247,3 +247,119 @@ def gaussian_nll_sum(loc, scale, y):
     if loc.is_cuda and loc.dtype == torch.float32:
         return _GaussLLFn.apply(loc.contiguous(), scale.contiguous(), y.contiguous())
     return _nll_ref(loc, scale, y)
+
+
+# --------------------------------------------------------------------------- #
+# Fused pre-activation depthwise conv block (1D).
+# Reference computation: npf/architectures/cnn.py ResConvBlock.forward
+# (reference cnn.py:204-215) — batchnorm -> relu -> depthwise conv1d
+# [-> + residual].  The pointwise conv that follows stays a library GEMM.
+# --------------------------------------------------------------------------- #
+
+
+def _conv_block_ref(x, weight, bias, bn, residual, training):
+    a = x
+    if bn is not None:
+        a = torch.nn.functional.batch_norm(
+            a, bn.running_mean, bn.running_var, bn.weight, bn.bias,
+            training, bn.momentum, bn.eps,
+        )
+    a = torch.relu(a)
+    out = torch.nn.functional.conv1d(
+        a, weight, bias, padding=weight.shape[-1] // 2, groups=x.shape[1]
+    )
+    if residual is not None:
+        out = out + residual
+    return out
+
+
+class _ConvBlock1dFn(torch.autograd.Function):
+    """bn+relu+dwconv(+residual) in 2 fwd / 2 bwd kernels (convblock.hip)."""
+
+    @staticmethod
+    def forward(ctx, x, weight, bias, gamma, beta, running_mean, running_var,
+                eps, momentum, training, residual):
+        ext = _backend.require_extension("conv_block_1d")
+        C = x.shape[1]
+        w2d = weight.view(C, -1).contiguous()
+        has_bn = gamma is not None
+        und = torch.Tensor()  # undefined-tensor placeholder for the binding
+        if has_bn:
+            if training:
+                mean, rstd, _ = ext.convblock_stats(
+                    x, running_mean if running_mean is not None else und,
+                    running_var if running_var is not None else und,
+                    eps, momentum,
+                )
+            else:
+                mean = running_mean
+                rstd = torch.rsqrt(running_var + eps)
+        else:
+            mean = rstd = und
+        y = ext.convblock_fwd(
+            x, residual if residual is not None else und, w2d,
+            bias if bias is not None else und,
+            gamma if has_bn else und, beta if has_bn else und, mean, rstd,
+        )
+        ctx.save_for_backward(x, w2d, *( (gamma, beta, mean, rstd) if has_bn else () ))
+        ctx.has_bn = has_bn
+        ctx.has_bias = bias is not None
+        ctx.has_res = residual is not None
+        ctx.training_mode = training
+        ctx.kshape = weight.shape
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        ext = _backend.require_extension("conv_block_1d")
+        und = torch.Tensor()
+        if ctx.has_bn:
+            x, w2d, gamma, beta, mean, rstd = ctx.saved_tensors
+        else:
+            x, w2d = ctx.saved_tensors
+            gamma = beta = mean = rstd = und
+        dy = dy.contiguous()
+        dx, dw, db, dgamma, dbeta = ext.convblock_bwd(
+            x, w2d, dy, gamma, beta, mean, rstd,
+            ctx.has_bias, ctx.training_mode,
+        )
+        return (
+            dx,
+            dw.view(ctx.kshape),
+            db if ctx.has_bias else None,
+            dgamma if ctx.has_bn else None,
+            dbeta if ctx.has_bn else None,
+            None, None, None, None, None,
+            dy if ctx.has_res else None,
+        )
+
+
+def conv_block_1d(x, conv, bn=None, residual=None):
+    """Fused norm->relu->depthwise-conv1d(+residual) on [N, C, L].
+
+    `conv` is the depthwise nn.Conv1d (groups == C), `bn` an optional
+    nn.BatchNorm1d (its running stats are updated in training mode exactly
+    like torch), `residual` an optional tensor added to the output.
+    """
+    if not x.is_cuda or _backend.require_extension("conv_block_1d") is None:
+        return _conv_block_ref(
+            x, conv.weight, conv.bias, bn,
+            residual, bn.training if bn is not None else conv.training,
+        )
+    training = bn.training if bn is not None else conv.training
+    # fp32 compute: BN statistics and the stencil are precision-sensitive;
+    # at these sizes the op is dispatch/HBM-bound so bf16 buys nothing
+    xf = x.float().contiguous()
+    res = residual.float().contiguous() if residual is not None else None
+    y = _ConvBlock1dFn.apply(
+        xf, conv.weight.float(), 
+        conv.bias.float() if conv.bias is not None else None,
+        bn.weight.float() if bn is not None else None,
+        bn.bias.float() if bn is not None else None,
+        bn.running_mean if bn is not None else None,
+        bn.running_var if bn is not None else None,
+        bn.eps if bn is not None else 1e-5,
+        bn.momentum if bn is not None else 0.1,
+        training, res,
+    )
+    return y

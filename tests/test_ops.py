@@ -177,3 +177,71 @@ class TestHIPKernels:
         ref.backward(dout.cpu())
         assert torch.allclose(loc.grad.cpu(), l0.grad, atol=1e-3)
         assert torch.allclose(scale.grad.cpu(), s0.grad, atol=1e-3)
+
+    @pytest.mark.parametrize("N,C,L,K,bn,res,training", [
+        (4, 8, 32, 5, True, True, True),
+        (32, 128, 192, 19, True, True, True),
+        (32, 128, 192, 19, True, False, True),
+        (16, 64, 100, 9, False, True, True),
+        (32, 128, 192, 19, True, True, False),
+        (512, 128, 192, 19, True, True, True),  # Z*B collapsed batch
+    ])
+    def test_conv_block_1d_fwd_bwd(self, N, C, L, K, bn, res, training):
+        import torch.nn as nn
+
+        from npf.ops.functional import _conv_block_ref
+
+        g = torch.Generator(device="cuda").manual_seed(0)
+        x = torch.randn(N, C, L, device="cuda", generator=g, requires_grad=True)
+        conv = nn.Conv1d(C, C, K, padding=K // 2, groups=C).cuda()
+        norm = nn.BatchNorm1d(C).cuda() if bn else None
+        if norm is not None:
+            with torch.no_grad():
+                norm.weight.uniform_(0.5, 1.5, generator=g)
+                norm.bias.uniform_(-0.2, 0.2, generator=g)
+                norm.running_mean.uniform_(-0.1, 0.1, generator=g)
+                norm.running_var.uniform_(0.5, 1.5, generator=g)
+            norm.train(training)
+        conv.train(training)
+        residual = (
+            torch.randn(N, C, L, device="cuda", generator=g, requires_grad=True)
+            if res else None
+        )
+
+        # fp32 oracle on CPU with cloned modules/buffers
+        conv0 = nn.Conv1d(C, C, K, padding=K // 2, groups=C)
+        conv0.load_state_dict({k: v.cpu() for k, v in conv.state_dict().items()})
+        norm0 = None
+        if norm is not None:
+            norm0 = nn.BatchNorm1d(C)
+            norm0.load_state_dict({k: v.cpu() for k, v in norm.state_dict().items()})
+            norm0.train(training)
+        x0 = x.detach().cpu().requires_grad_(True)
+        r0 = residual.detach().cpu().requires_grad_(True) if res else None
+
+        out = F_ops.conv_block_1d(x, conv, bn=norm, residual=residual)
+        ref = _conv_block_ref(x0, conv0.weight, conv0.bias, norm0, r0, training)
+        assert torch.allclose(out.cpu(), ref, atol=2e-4), (
+            (out.cpu() - ref).abs().max()
+        )
+
+        dout = torch.randn_like(out)
+        out.backward(dout)
+        ref.backward(dout.cpu())
+        pairs = [(x.grad, x0.grad), (conv.weight.grad, conv0.weight.grad),
+                 (conv.bias.grad, conv0.bias.grad)]
+        if res:
+            pairs.append((residual.grad, r0.grad))
+        if bn:
+            pairs += [(norm.weight.grad, norm0.weight.grad),
+                      (norm.bias.grad, norm0.bias.grad)]
+        for a, b in pairs:
+            assert torch.allclose(a.cpu(), b, atol=3e-3), (a.cpu() - b).abs().max()
+        if bn and training:
+            # running-stat update parity (in-place on the GPU module)
+            assert torch.allclose(
+                norm.running_mean.cpu(), norm0.running_mean, atol=1e-4
+            )
+            assert torch.allclose(
+                norm.running_var.cpu(), norm0.running_var, atol=1e-4
+            )
