@@ -292,3 +292,17 @@ extern "C" void npf_cb_bwd_launch(const float* x, const float* w,
                      stream, x, dact, dy, gamma, mean, rstd, sum_dxhat,
                      sum_dxhat_xhat, dx, N, C, L, training);
 }
+
+// standalone launcher for the elementwise BN-backward kernel (shared by the
+// 2D block, which flattens L = H*W)
+extern "C" void npf_cb_bwd_dx_launch(const float* x, const float* dact,
+                                     const float* dy, const float* gamma,
+                                     const float* mean, const float* rstd,
+                                     const float* sum_dxhat,
+                                     const float* sum_dxhat_xhat, float* dx,
+                                     int N, int C, int L, int training,
+                                     hipStream_t stream) {
+  hipLaunchKernelGGL(npf_cb_bwd_dx, dim3((unsigned)N * C), dim3(CB_BLOCK), 0,
+                     stream, x, dact, dy, gamma, mean, rstd, sum_dxhat,
+                     sum_dxhat_xhat, dx, N, C, L, training);
+}

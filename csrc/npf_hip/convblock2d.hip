@@ -1,0 +1,205 @@
+// Fused pre-activation depthwise conv block (2D) — the GridConvCNP/LNP CNN
+// hot loop (reference cnn.py:204-215 with Conv2d/BatchNorm2d, k=9, on
+// [N, C, H, W] = [Z*B, 128, 32..64, 32..64]).
+//
+//   a = relu(batchnorm(x));  y = dwconv2d(a) (+bias) (+res)
+//
+// Same structure as the 1D kernels (convblock.hip): the per-channel stats
+// and the elementwise BN-backward kernel are shared with 1D (L = H*W);
+// only the two stencil kernels differ.  One workgroup per (n, c) plane,
+// 256 threads striding H*W; the activation plane is staged in LDS with its
+// (k/2)-halo ((64+8)^2 floats = 20 KB of the 160 KB LDS), and the k*k
+// weights live in LDS (k^2 register accumulators would blow VGPR budget);
+// dW is accumulated as one block-reduction per weight position, reading
+// both operands from LDS.
+
+#include "common.h"
+
+#define CB2_BLOCK 256
+#define CB2_MAX_K 13
+
+extern "C" __global__ void __launch_bounds__(CB2_BLOCK)
+npf_cb2d_fwd(const float* __restrict__ x, const float* __restrict__ res,
+             const float* __restrict__ w, const float* __restrict__ bias,
+             const float* __restrict__ gamma, const float* __restrict__ beta,
+             const float* __restrict__ mean, const float* __restrict__ rstd,
+             float* __restrict__ y, int N, int C, int H, int W, int K) {
+  extern __shared__ float smem[];
+  const int pad = K / 2;
+  const int HP = H + 2 * pad, WP = W + 2 * pad;
+  float* a = smem;            // [HP * WP] activation plane with halo
+  float* ws = smem + HP * WP; // [K * K]
+
+  const int n = blockIdx.x / C;
+  const int c = blockIdx.x % C;
+  const float* xpl = x + ((long)n * C + c) * H * W;
+
+  const bool has_bn = gamma != nullptr;
+  const float mu = has_bn ? mean[c] : 0.f;
+  const float gscale = has_bn ? rstd[c] * gamma[c] : 1.f;
+  const float gshift = has_bn ? beta[c] : 0.f;
+
+  for (int i = threadIdx.x; i < K * K; i += CB2_BLOCK) ws[i] = w[c * K * K + i];
+  for (int i = threadIdx.x; i < HP * WP; i += CB2_BLOCK) {
+    const int r = i / WP - pad, col = i % WP - pad;
+    float v = 0.f;
+    if (r >= 0 && r < H && col >= 0 && col < W)
+      v = fmaxf((xpl[r * W + col] - mu) * gscale + gshift, 0.f);
+    a[i] = v;
+  }
+  __syncthreads();
+
+  const float b = (bias != nullptr) ? bias[c] : 0.f;
+  float* ypl = y + ((long)n * C + c) * H * W;
+  const float* rpl =
+      (res != nullptr) ? res + ((long)n * C + c) * H * W : nullptr;
+  for (int i = threadIdx.x; i < H * W; i += CB2_BLOCK) {
+    const int r = i / W, col = i % W;
+    float acc = b;
+    for (int kr = 0; kr < K; ++kr) {
+      const float* arow = a + (r + kr) * WP + col;
+      #pragma unroll 3
+      for (int kc = 0; kc < K; ++kc) acc += ws[kr * K + kc] * arow[kc];
+    }
+    if (rpl != nullptr) acc += rpl[i];
+    ypl[i] = acc;
+  }
+}
+
+// backward stencil: dact (stored), dW/db block-reduced then global atomics,
+// BN channel partial sums (global atomics)
+extern "C" __global__ void __launch_bounds__(CB2_BLOCK)
+npf_cb2d_bwd_dact(const float* __restrict__ x, const float* __restrict__ w,
+                  const float* __restrict__ dy, const float* __restrict__ gamma,
+                  const float* __restrict__ beta, const float* __restrict__ mean,
+                  const float* __restrict__ rstd, float* __restrict__ dact,
+                  float* __restrict__ dw, float* __restrict__ db,
+                  float* __restrict__ sum_dxhat,
+                  float* __restrict__ sum_dxhat_xhat,
+                  float* __restrict__ dgamma, float* __restrict__ dbeta,
+                  int N, int C, int H, int W, int K) {
+  extern __shared__ float smem[];
+  const int pad = K / 2;
+  const int HP = H + 2 * pad, WP = W + 2 * pad;
+  float* a = smem;                     // [HP*WP] activations (halo)
+  float* dys = smem + HP * WP;         // [HP*WP] dY (halo)
+  float* ws = dys + HP * WP;           // [K*K]
+  __shared__ float red[16];
+
+  const int n = blockIdx.x / C;
+  const int c = blockIdx.x % C;
+  const float* xpl = x + ((long)n * C + c) * H * W;
+  const float* dypl = dy + ((long)n * C + c) * H * W;
+
+  const bool has_bn = gamma != nullptr;
+  const float mu = has_bn ? mean[c] : 0.f;
+  const float rs = has_bn ? rstd[c] : 1.f;
+  const float gm = has_bn ? gamma[c] : 1.f;
+  const float gscale = has_bn ? rs * gm : 1.f;
+  const float gshift = has_bn ? beta[c] : 0.f;
+
+  for (int i = threadIdx.x; i < K * K; i += CB2_BLOCK) ws[i] = w[c * K * K + i];
+  for (int i = threadIdx.x; i < HP * WP; i += CB2_BLOCK) {
+    const int r = i / WP - pad, col = i % WP - pad;
+    const bool in = (r >= 0 && r < H && col >= 0 && col < W);
+    dys[i] = in ? dypl[r * W + col] : 0.f;
+    float v = 0.f;
+    if (in) v = fmaxf((xpl[r * W + col] - mu) * gscale + gshift, 0.f);
+    a[i] = v;
+  }
+  __syncthreads();
+
+  float* dactpl = dact + ((long)n * C + c) * H * W;
+  float s_dxhat = 0.f, s_dxhat_xhat = 0.f, s_dg = 0.f, s_db = 0.f, dbp = 0.f;
+  for (int i = threadIdx.x; i < H * W; i += CB2_BLOCK) {
+    const int r = i / W, col = i % W;
+    const float dyl = dypl[i];
+    dbp += dyl;
+    // transposed conv: da[p] = sum_k w[k] dY[p + pad - k]
+    float da = 0.f;
+    for (int kr = 0; kr < K; ++kr) {
+      const float* drow = dys + (r + K - 1 - kr) * WP + col;
+      #pragma unroll 3
+      for (int kc = 0; kc < K; ++kc) da += ws[kr * K + kc] * drow[K - 1 - kc];
+    }
+    const float act = a[(r + pad) * WP + col + pad];
+    const float dr = (act > 0.f) ? da : 0.f;
+    dactpl[i] = dr;
+    if (has_bn) {
+      const float xhat = (xpl[i] - mu) * rs;
+      const float dxh = dr * gm;
+      s_dxhat += dxh;
+      s_dxhat_xhat += dxh * xhat;
+      s_dg += dr * xhat;
+      s_db += dr;
+    }
+  }
+  __syncthreads();
+  // dW[kr,kc] = sum_p dY[p] * a[p + (kr,kc)]: one block-reduction per weight
+  // position, every operand read from LDS (K^2 register accumulators per
+  // thread would blow the VGPR budget; per-pixel LDS atomics would serialize)
+  for (int kk = 0; kk < K * K; ++kk) {
+    const int kr = kk / K, kc = kk % K;
+    float psum = 0.f;
+    for (int i = threadIdx.x; i < H * W; i += CB2_BLOCK) {
+      const int r = i / W, col = i % W;
+      psum += dys[(r + pad) * WP + col + pad] * a[(r + kr) * WP + col + kc];
+    }
+    psum = block_reduce_sum(psum, red);
+    __syncthreads();
+    if (threadIdx.x == 0) atomicAdd(&dw[c * K * K + kk], psum);
+  }
+  {
+    const float v = block_reduce_sum(dbp, red);
+    __syncthreads();
+    if (threadIdx.x == 0 && db != nullptr) atomicAdd(&db[c], v);
+  }
+  if (has_bn) {
+    float v = block_reduce_sum(s_dxhat, red);
+    __syncthreads();
+    if (threadIdx.x == 0) atomicAdd(&sum_dxhat[c], v);
+    v = block_reduce_sum(s_dxhat_xhat, red);
+    __syncthreads();
+    if (threadIdx.x == 0) atomicAdd(&sum_dxhat_xhat[c], v);
+    v = block_reduce_sum(s_dg, red);
+    __syncthreads();
+    if (threadIdx.x == 0) atomicAdd(&dgamma[c], v);
+    v = block_reduce_sum(s_db, red);
+    __syncthreads();
+    if (threadIdx.x == 0) atomicAdd(&dbeta[c], v);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// host launchers (stats + bn-backward-elementwise reuse the 1D kernels with
+// L = H*W, launched from ext.cpp)
+// ---------------------------------------------------------------------------
+
+extern "C" void npf_cb2d_fwd_launch(const float* x, const float* res,
+                                    const float* w, const float* bias,
+                                    const float* gamma, const float* beta,
+                                    const float* mean, const float* rstd,
+                                    float* y, int N, int C, int H, int W,
+                                    int K, hipStream_t stream) {
+  const int pad = K / 2;
+  const size_t smem =
+      ((size_t)(H + 2 * pad) * (W + 2 * pad) + K * K) * sizeof(float);
+  hipLaunchKernelGGL(npf_cb2d_fwd, dim3((unsigned)N * C), dim3(CB2_BLOCK),
+                     smem, stream, x, res, w, bias, gamma, beta, mean, rstd, y,
+                     N, C, H, W, K);
+}
+
+extern "C" void npf_cb2d_bwd_dact_launch(
+    const float* x, const float* w, const float* dy, const float* gamma,
+    const float* beta, const float* mean, const float* rstd, float* dact,
+    float* dw, float* db, float* sum_dxhat, float* sum_dxhat_xhat,
+    float* dgamma, float* dbeta, int N, int C, int H, int W, int K,
+    hipStream_t stream) {
+  const int pad = K / 2;
+  const size_t smem =
+      (2 * (size_t)(H + 2 * pad) * (W + 2 * pad) + K * K) * sizeof(float);
+  hipLaunchKernelGGL(npf_cb2d_bwd_dact, dim3((unsigned)N * C),
+                     dim3(CB2_BLOCK), smem, stream, x, w, dy, gamma, beta,
+                     mean, rstd, dact, dw, db, sum_dxhat, sum_dxhat_xhat,
+                     dgamma, dbeta, N, C, H, W, K);
+}

@@ -47,6 +47,19 @@ void npf_cb_bwd_launch(const float*, const float*, const float*, const float*,
                        const float*, const float*, const float*, float*,
                        float*, float*, float*, float*, float*, float*, float*,
                        int, int, int, int, int, hipStream_t);
+void npf_cb_bwd_dx_launch(const float*, const float*, const float*,
+                          const float*, const float*, const float*,
+                          const float*, const float*, float*, int, int, int,
+                          int, hipStream_t);
+void npf_cb2d_fwd_launch(const float*, const float*, const float*,
+                         const float*, const float*, const float*,
+                         const float*, const float*, float*, int, int, int,
+                         int, int, hipStream_t);
+void npf_cb2d_bwd_dact_launch(const float*, const float*, const float*,
+                              const float*, const float*, const float*,
+                              const float*, float*, float*, float*, float*,
+                              float*, float*, float*, int, int, int, int, int,
+                              hipStream_t);
 }
 
 namespace {
@@ -274,6 +287,77 @@ std::vector<torch::Tensor> convblock_bwd(torch::Tensor x, torch::Tensor w,
   return {dx, dw, db, dgamma, dbeta};
 }
 
+
+// 2D fused conv block (GridConv models); stats reuse convblock_stats with
+// x viewed as [N, C, H*W]
+torch::Tensor convblock2d_fwd(torch::Tensor x, torch::Tensor res,
+                              torch::Tensor w, torch::Tensor b,
+                              torch::Tensor gamma, torch::Tensor beta,
+                              torch::Tensor mean, torch::Tensor rstd) {
+  check_cuda_contig(x, "x");
+  check_cuda_contig(w, "w");
+  TORCH_CHECK(x.scalar_type() == torch::kFloat32, "convblock is fp32");
+  const int N = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
+  const int K = w.size(-1);
+  TORCH_CHECK(K <= 13 && K % 2 == 1, "2D kernel size must be odd and <= 13");
+  auto y = torch::empty_like(x);
+  auto stream = at::hip::getCurrentHIPStream();
+  npf_cb2d_fwd_launch(
+      x.data_ptr<float>(), res.defined() ? res.data_ptr<float>() : nullptr,
+      w.data_ptr<float>(), b.defined() ? b.data_ptr<float>() : nullptr,
+      gamma.defined() ? gamma.data_ptr<float>() : nullptr,
+      gamma.defined() ? beta.data_ptr<float>() : nullptr,
+      gamma.defined() ? mean.data_ptr<float>() : nullptr,
+      gamma.defined() ? rstd.data_ptr<float>() : nullptr, y.data_ptr<float>(),
+      N, C, H, W, K, stream);
+  return y;
+}
+
+std::vector<torch::Tensor> convblock2d_bwd(torch::Tensor x, torch::Tensor w,
+                                           torch::Tensor dy,
+                                           torch::Tensor gamma,
+                                           torch::Tensor beta,
+                                           torch::Tensor mean,
+                                           torch::Tensor rstd, bool has_bias,
+                                           bool training) {
+  check_cuda_contig(x, "x");
+  check_cuda_contig(w, "w");
+  check_cuda_contig(dy, "dy");
+  const int N = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
+  const int K = w.size(-1);
+  const bool has_bn = gamma.defined();
+  auto dact = torch::empty_like(x);
+  auto dx = torch::empty_like(x);
+  auto dw = torch::zeros_like(w);
+  auto db = has_bias ? torch::zeros({C}, x.options()) : torch::Tensor();
+  auto opts = x.options();
+  auto sum_dxhat = has_bn ? torch::zeros({C}, opts) : torch::Tensor();
+  auto sum_dxhat_xhat = has_bn ? torch::zeros({C}, opts) : torch::Tensor();
+  auto dgamma = has_bn ? torch::zeros({C}, opts) : torch::Tensor();
+  auto dbeta = has_bn ? torch::zeros({C}, opts) : torch::Tensor();
+  auto stream = at::hip::getCurrentHIPStream();
+  npf_cb2d_bwd_dact_launch(
+      x.data_ptr<float>(), w.data_ptr<float>(), dy.data_ptr<float>(),
+      has_bn ? gamma.data_ptr<float>() : nullptr,
+      has_bn ? beta.data_ptr<float>() : nullptr,
+      has_bn ? mean.data_ptr<float>() : nullptr,
+      has_bn ? rstd.data_ptr<float>() : nullptr, dact.data_ptr<float>(),
+      dw.data_ptr<float>(), has_bias ? db.data_ptr<float>() : nullptr,
+      has_bn ? sum_dxhat.data_ptr<float>() : nullptr,
+      has_bn ? sum_dxhat_xhat.data_ptr<float>() : nullptr,
+      has_bn ? dgamma.data_ptr<float>() : nullptr,
+      has_bn ? dbeta.data_ptr<float>() : nullptr, N, C, H, W, K, stream);
+  npf_cb_bwd_dx_launch(
+      x.data_ptr<float>(), dact.data_ptr<float>(), dy.data_ptr<float>(),
+      has_bn ? gamma.data_ptr<float>() : nullptr,
+      has_bn ? mean.data_ptr<float>() : nullptr,
+      has_bn ? rstd.data_ptr<float>() : nullptr,
+      has_bn ? sum_dxhat.data_ptr<float>() : nullptr,
+      has_bn ? sum_dxhat_xhat.data_ptr<float>() : nullptr,
+      dx.data_ptr<float>(), N, C, H * W, training ? 1 : 0, stream);
+  return {dx, dw, db, dgamma, dbeta};
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -289,5 +373,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "fused bn+relu+depthwise-conv(+residual) forward, 1D");
   m.def("convblock_bwd", &convblock_bwd,
         "fused conv block backward -> (dx, dw, db, dgamma, dbeta)");
+  m.def("convblock2d_fwd", &convblock2d_fwd,
+        "fused bn+relu+depthwise-conv(+residual) forward, 2D");
+  m.def("convblock2d_bwd", &convblock2d_bwd,
+        "fused 2D conv block backward -> (dx, dw, db, dgamma, dbeta)");
   m.attr("_arch") = "gfx950";
 }

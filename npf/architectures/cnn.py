@@ -129,37 +129,39 @@ class ResConvBlock(nn.Module):
     def reset_parameters(self):
         weights_init(self)
 
-    def _is_fusable_1d(self, X):
-        from npf.ops import has_extension
+    def _fused_op(self, X):
+        from npf.ops import conv_block_1d, conv_block_2d, has_extension
 
-        return (
-            X.is_cuda
-            and X.dim() == 3
-            and isinstance(self.conv2_depthwise, nn.Conv1d)
-            and isinstance(self.activation, nn.ReLU)
-            and isinstance(self.norm2, (nn.BatchNorm1d, nn.Identity))
-            and has_extension()
-        )
+        if not (X.is_cuda and isinstance(self.activation, nn.ReLU)
+                and has_extension()):
+            return None
+        if (X.dim() == 3 and isinstance(self.conv2_depthwise, nn.Conv1d)
+                and isinstance(self.norm2, (nn.BatchNorm1d, nn.Identity))):
+            return conv_block_1d
+        if (X.dim() == 4 and isinstance(self.conv2_depthwise, nn.Conv2d)
+                and isinstance(self.norm2, (nn.BatchNorm2d, nn.Identity))
+                and self.conv2_depthwise.padding_mode == "zeros"):
+            return conv_block_2d
+        return None
 
     def forward(self, X):
-        if self._is_fusable_1d(X):
-            # fused HIP path (csrc/npf_hip/convblock.hip): bn+relu+dwconv
+        fused = self._fused_op(X)
+        if fused is not None:
+            # fused HIP path (csrc/npf_hip/convblock{,2d}.hip): bn+relu+dwconv
             # collapse to 2 kernels; the pointwise stays a library GEMM
-            from npf.ops import conv_block_1d
-
             if self.n_conv_layers == 2:
-                h = conv_block_1d(
+                h = fused(
                     X,
                     self.conv1.depthwise,
-                    bn=self.norm1 if isinstance(self.norm1, nn.BatchNorm1d) else None,
+                    bn=self.norm1 if not isinstance(self.norm1, nn.Identity) else None,
                 )
                 h = self.conv1.pointwise(h)
             else:
                 h = X
-            out = conv_block_1d(
+            out = fused(
                 h,
                 self.conv2_depthwise,
-                bn=self.norm2 if isinstance(self.norm2, nn.BatchNorm1d) else None,
+                bn=self.norm2 if not isinstance(self.norm2, nn.Identity) else None,
                 residual=X,
             )
             return self.conv2_pointwise(out)

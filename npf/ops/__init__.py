@@ -1,12 +1,13 @@
 """MI355X ops layer: fused CDNA4 HIP kernels with torch reference fallbacks."""
 
 from ._backend import extension, has_extension, require_extension
-from .functional import (attention_qkv, conv_block_1d, gaussian_nll_sum,
-                         setconv_gaussian)
+from .functional import (attention_qkv, conv_block_1d, conv_block_2d,
+                         gaussian_nll_sum, setconv_gaussian)
 
 __all__ = [
     "attention_qkv",
     "conv_block_1d",
+    "conv_block_2d",
     "setconv_gaussian",
     "gaussian_nll_sum",
     "extension",

@@ -12,7 +12,7 @@ import torch
 
 from . import _backend
 
-__all__ = ["attention_qkv", "setconv_gaussian", "gaussian_nll_sum", "conv_block_1d"]
+__all__ = ["attention_qkv", "setconv_gaussian", "gaussian_nll_sum", "conv_block_1d", "conv_block_2d"]
 
 
 # --------------------------------------------------------------------------- #
@@ -363,3 +363,120 @@ def conv_block_1d(x, conv, bn=None, residual=None):
         training, res,
     )
     return y
+
+
+# --------------------------------------------------------------------------- #
+# Fused pre-activation depthwise conv block (2D) — GridConv models.
+# --------------------------------------------------------------------------- #
+
+
+def _conv_block2d_ref(x, weight, bias, bn, residual, training):
+    a = x
+    if bn is not None:
+        a = torch.nn.functional.batch_norm(
+            a, bn.running_mean, bn.running_var, bn.weight, bn.bias,
+            training, bn.momentum, bn.eps,
+        )
+    a = torch.relu(a)
+    out = torch.nn.functional.conv2d(
+        a, weight, bias, padding=weight.shape[-1] // 2, groups=x.shape[1]
+    )
+    if residual is not None:
+        out = out + residual
+    return out
+
+
+class _ConvBlock2dFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, weight, bias, gamma, beta, running_mean, running_var,
+                eps, momentum, training, residual):
+        ext = _backend.require_extension("conv_block_2d")
+        C = x.shape[1]
+        K = weight.shape[-1]
+        w2d = weight.view(C, K * K).contiguous()
+        has_bn = gamma is not None
+        und = torch.Tensor()
+        if has_bn:
+            if training:
+                mean, rstd, _ = ext.convblock_stats(
+                    x.view(x.shape[0], C, -1),
+                    running_mean if running_mean is not None else und,
+                    running_var if running_var is not None else und,
+                    eps, momentum,
+                )
+            else:
+                mean = running_mean
+                rstd = torch.rsqrt(running_var + eps)
+        else:
+            mean = rstd = und
+        y = ext.convblock2d_fwd(
+            x, residual if residual is not None else und, w2d,
+            bias if bias is not None else und,
+            gamma if has_bn else und, beta if has_bn else und, mean, rstd,
+        )
+        ctx.save_for_backward(x, w2d, *( (gamma, beta, mean, rstd) if has_bn else () ))
+        ctx.has_bn = has_bn
+        ctx.has_bias = bias is not None
+        ctx.has_res = residual is not None
+        ctx.training_mode = training
+        ctx.kshape = weight.shape
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        ext = _backend.require_extension("conv_block_2d")
+        und = torch.Tensor()
+        if ctx.has_bn:
+            x, w2d, gamma, beta, mean, rstd = ctx.saved_tensors
+        else:
+            x, w2d = ctx.saved_tensors
+            gamma = beta = mean = rstd = und
+        dy = dy.contiguous()
+        dx, dw, db, dgamma, dbeta = ext.convblock2d_bwd(
+            x, w2d, dy, gamma, beta, mean, rstd,
+            ctx.has_bias, ctx.training_mode,
+        )
+        return (
+            dx,
+            dw.view(ctx.kshape),
+            db if ctx.has_bias else None,
+            dgamma if ctx.has_bn else None,
+            dbeta if ctx.has_bn else None,
+            None, None, None, None, None,
+            dy if ctx.has_res else None,
+        )
+
+
+def _cb2d_lds_ok(x, k):
+    """The 2D backward stages two haloed planes in LDS; stay under the 64 KB
+    default dynamic-LDS limit, else fall back to the composed path."""
+    H, W = x.shape[-2], x.shape[-1]
+    pad = k // 2
+    return 2 * (H + 2 * pad) * (W + 2 * pad) * 4 <= 64 * 1024
+
+
+def conv_block_2d(x, conv, bn=None, residual=None):
+    """Fused norm->relu->depthwise-conv2d(+residual) on [N, C, H, W]."""
+    training = bn.training if bn is not None else conv.training
+    k = conv.weight.shape[-1]
+    if (
+        not x.is_cuda
+        or not _cb2d_lds_ok(x, k)
+        or _backend.require_extension("conv_block_2d") is None
+    ):
+        return _conv_block2d_ref(
+            x, conv.weight, conv.bias, bn, residual, training
+        )
+    xf = x.float().contiguous()
+    res = residual.float().contiguous() if residual is not None else None
+    return _ConvBlock2dFn.apply(
+        xf, conv.weight.float(),
+        conv.bias.float() if conv.bias is not None else None,
+        bn.weight.float() if bn is not None else None,
+        bn.bias.float() if bn is not None else None,
+        bn.running_mean if bn is not None else None,
+        bn.running_var if bn is not None else None,
+        bn.eps if bn is not None else 1e-5,
+        bn.momentum if bn is not None else 0.1,
+        training, res,
+    )
