@@ -25,6 +25,7 @@ def build():
         os.path.join(REPO, "csrc", "npf_hip", "gauss_ll.hip"),
         os.path.join(REPO, "csrc", "npf_hip", "convblock.hip"),
         os.path.join(REPO, "csrc", "npf_hip", "convblock2d.hip"),
+        os.path.join(REPO, "csrc", "npf_hip", "griddensity.hip"),
     ]
     ext = CUDAExtension(
         name="npf._hip_C",

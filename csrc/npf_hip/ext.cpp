@@ -60,6 +60,11 @@ void npf_cb2d_bwd_dact_launch(const float*, const float*, const float*,
                               const float*, float*, float*, float*, float*,
                               float*, float*, float*, int, int, int, int, int,
                               hipStream_t);
+void npf_gde_fwd_launch(const float*, const float*, const float*, float*, int,
+                        int, int, int, int, hipStream_t);
+void npf_gde_bwd_launch(const float*, const float*, const float*, const float*,
+                        const float*, float*, float*, int, int, int, int, int,
+                        hipStream_t);
 }
 
 namespace {
@@ -358,6 +363,44 @@ std::vector<torch::Tensor> convblock2d_bwd(torch::Tensor x, torch::Tensor w,
   return {dx, dw, db, dgamma, dbeta};
 }
 
+
+// grid density encoder (GridConv masked abs-conv pair + divide + concat)
+torch::Tensor griddensity_fwd(torch::Tensor x, torch::Tensor m,
+                              torch::Tensor w) {
+  check_cuda_contig(x, "x");
+  check_cuda_contig(m, "m");
+  check_cuda_contig(w, "w");
+  TORCH_CHECK(x.scalar_type() == torch::kFloat32, "griddensity is fp32");
+  const int B = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
+  const int K = w.size(-1);
+  TORCH_CHECK(K <= 13 && K % 2 == 1, "kernel size must be odd and <= 13");
+  auto out = torch::empty({B, 2 * C, H, W}, x.options());
+  auto stream = at::hip::getCurrentHIPStream();
+  npf_gde_fwd_launch(x.data_ptr<float>(), m.data_ptr<float>(),
+                     w.data_ptr<float>(), out.data_ptr<float>(), B, C, H, W,
+                     K, stream);
+  return out;
+}
+
+std::tuple<torch::Tensor, torch::Tensor> griddensity_bwd(torch::Tensor x,
+                                                         torch::Tensor m,
+                                                         torch::Tensor dout,
+                                                         torch::Tensor out,
+                                                         torch::Tensor w) {
+  check_cuda_contig(dout, "dout");
+  check_cuda_contig(out, "out");
+  const int B = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
+  const int K = w.size(-1);
+  auto dx = torch::empty_like(x);
+  auto dw = torch::empty_like(w);
+  auto stream = at::hip::getCurrentHIPStream();
+  npf_gde_bwd_launch(x.data_ptr<float>(), m.data_ptr<float>(),
+                     dout.data_ptr<float>(), out.data_ptr<float>(),
+                     w.data_ptr<float>(), dx.data_ptr<float>(),
+                     dw.data_ptr<float>(), B, C, H, W, K, stream);
+  return {dx, dw};
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -377,5 +420,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "fused bn+relu+depthwise-conv(+residual) forward, 2D");
   m.def("convblock2d_bwd", &convblock2d_bwd,
         "fused 2D conv block backward -> (dx, dw, db, dgamma, dbeta)");
+  m.def("griddensity_fwd", &griddensity_fwd,
+        "fused masked abs-conv density encoder forward");
+  m.def("griddensity_bwd", &griddensity_bwd,
+        "fused density encoder backward -> (dx, dw)");
   m.attr("_arch") = "gfx950";
 }

@@ -79,6 +79,21 @@ class GridConvCNP(NeuralProcessFamily):
 
     dflt_Modules = ConvCNP.dflt_Modules
 
+    def _fused_density_ok(self, X):
+        from npf.ops import has_extension
+
+        conv = self.conv
+        return (
+            X.is_cuda
+            and has_extension()
+            and isinstance(conv, nn.Conv2d)
+            and conv.bias is None
+            and conv.groups == conv.in_channels
+            and conv.padding_mode == "zeros"
+            and conv.dilation == (1, 1)
+            and "AbsConv" in type(conv).__name__
+        )
+
     def cntxt_to_induced(self, mask_cntxt, X):
         """Masked abs-depthwise conv pair -> normalized signal + density
         (reference gridconvnp.py:136-162)."""
@@ -86,11 +101,18 @@ class GridConvCNP(NeuralProcessFamily):
         X = channels_to_2nd_dim(X)
         mask_cntxt = channels_to_2nd_dim(mask_cntxt).float()
 
-        signal = self.conv(X * mask_cntxt)
-        density = self.conv(mask_cntxt.expand_as(X))
-        out = signal / torch.clamp(density, min=1e-5)
+        if self._fused_density_ok(X):
+            # ONE HIP kernel (csrc/npf_hip/griddensity.hip) for the masked
+            # abs-conv pair + divide + concat
+            from npf.ops import grid_density
 
-        out = torch.cat([out, density], dim=1)
+            out = grid_density(X, mask_cntxt, self.conv.weight)
+        else:
+            signal = self.conv(X * mask_cntxt)
+            density = self.conv(mask_cntxt.expand_as(X))
+            out = signal / torch.clamp(density, min=1e-5)
+            out = torch.cat([out, density], dim=1)
+
         out = channels_to_last_dim(out)
         return self.resizer(out)  # [B, *grid, r_dim]
 

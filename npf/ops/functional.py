@@ -12,7 +12,7 @@ import torch
 
 from . import _backend
 
-__all__ = ["attention_qkv", "setconv_gaussian", "gaussian_nll_sum", "conv_block_1d", "conv_block_2d"]
+__all__ = ["attention_qkv", "setconv_gaussian", "gaussian_nll_sum", "conv_block_1d", "conv_block_2d", "grid_density"]
 
 
 # --------------------------------------------------------------------------- #
@@ -479,4 +479,55 @@ def conv_block_2d(x, conv, bn=None, residual=None):
         bn.eps if bn is not None else 1e-5,
         bn.momentum if bn is not None else 0.1,
         training, res,
+    )
+
+
+# --------------------------------------------------------------------------- #
+# Grid density encoder (GridConvCNP/LNP cntxt_to_induced).
+# Reference computation: gridconvnp.py:136-162 — abs-weight depthwise conv on
+# the masked image and on the mask, divide (clamp 1e-5), concat density.
+# --------------------------------------------------------------------------- #
+
+
+def _grid_density_ref(x, mask, weight):
+    w = weight.abs()
+    C = x.shape[1]
+    pad = weight.shape[-1] // 2
+    signal = torch.nn.functional.conv2d(x * mask, w, padding=pad, groups=C)
+    density = torch.nn.functional.conv2d(
+        mask.expand_as(x), w, padding=pad, groups=C
+    )
+    out = signal / torch.clamp(density, min=1e-5)
+    return torch.cat([out, density], dim=1)
+
+
+class _GridDensityFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, mask, weight):
+        ext = _backend.require_extension("grid_density")
+        C = x.shape[1]
+        K = weight.shape[-1]
+        w2d = weight.view(C, K * K).contiguous()
+        out = ext.griddensity_fwd(x, mask, w2d)
+        ctx.save_for_backward(x, mask, w2d, out)
+        ctx.kshape = weight.shape
+        return out
+
+    @staticmethod
+    def backward(ctx, dout):
+        ext = _backend.require_extension("grid_density")
+        x, mask, w2d, out = ctx.saved_tensors
+        dx, dw = ext.griddensity_bwd(x, mask, dout.contiguous(), out, w2d)
+        return dx, None, dw.view(ctx.kshape)
+
+
+def grid_density(x, mask, weight):
+    """Fused density encoder: x [B,C,H,W], mask [B,C or 1,H,W] (no grad),
+    abs-conv weight [C,1,K,K] -> [B, 2C, H, W] (normalized signal ; density).
+    """
+    if not x.is_cuda or _backend.require_extension("grid_density") is None:
+        return _grid_density_ref(x, mask, weight)
+    mask = mask.float().expand_as(x).contiguous()
+    return _GridDensityFn.apply(
+        x.float().contiguous(), mask, weight.float()
     )
