@@ -50,6 +50,8 @@ __all__ = [
     "plot_multi_posterior_samples_1d",
     "plot_multi_posterior_samples_imgs",
     "plot_multi_prior_samples_1d",
+    "gif_explain",
+    "select_labels",
 ]
 
 
@@ -416,3 +418,170 @@ def plot_multi_prior_samples_1d(trainers, datasets, **kwargs):
         )
     plt.tight_layout()
     return fig
+
+
+# --------------------------------------------------------------------------- #
+# ConvCNP anatomy gif (reference ntbks_helpers.py:485-720 gif_explain)
+# --------------------------------------------------------------------------- #
+
+
+def select_labels(dataset, label):
+    """Subset an image dataset to one class label."""
+    import torch as _torch
+
+    dataset = copy.deepcopy(dataset)
+    targets = (
+        dataset.targets
+        if _torch.is_tensor(dataset.targets)
+        else _torch.as_tensor(dataset.targets)
+    )
+    filt = targets == label
+    dataset.data = dataset.data[filt]
+    dataset.targets = targets[filt]
+    return dataset
+
+
+def _convcnp_stages(model, X_cntxt, Y_cntxt, X_trgt):
+    """Run a ConvCNP forward stage by stage, returning the intermediate
+    representations the anatomy gif plots (reference's `splitted_forward`):
+    pre-resizer SetConv output (values + density channel), the post-CNN
+    induced representation, the target representation, and the predictive."""
+    import torch as _torch
+
+    X_cntxt_e = model.x_encoder(X_cntxt)
+    X_trgt_e = model.x_encoder(X_trgt)
+    X_induced = model._get_X_induced(X_cntxt_e)
+
+    resizer = model.cntxt_to_induced.resizer
+    model.cntxt_to_induced.resizer = _torch.nn.Identity()
+    try:
+        R_setconv = model.cntxt_to_induced(X_cntxt_e, X_induced, Y_cntxt)
+    finally:
+        model.cntxt_to_induced.resizer = resizer
+
+    R = model.encode_globally(X_cntxt_e, Y_cntxt)
+    R_trgt = model.trgt_dependent_representation(X_cntxt_e, None, R, X_trgt_e)
+    p_yCc = model.decode(X_trgt_e, R_trgt)
+    return X_induced, R_setconv, R, R_trgt, p_yCc
+
+
+def gif_explain(
+    save_filename,
+    dataset,
+    model,
+    plot_config_kwargs=dict(),
+    seed=123,
+    n_cntxt=10,
+    fps=0.5,
+    length_scale_delta=0,
+):
+    """Step-by-step ConvCNP anatomy gif: context points -> SetConv values ->
+    density channel -> post-CNN channels -> target representation ->
+    predictive distribution."""
+    import torch as _torch
+
+    from npf.train.helpers import set_seed
+    from npf.viz.helpers import fig2img
+    from npf.viz.viz_1d import _plot_posterior_predefined_cntxt
+
+    set_seed(seed)
+    X, Y = dataset.get_samples(n_samples=1, n_points=dataset.n_points)
+    X_cntxt, Y_cntxt, X_trgt, Y_trgt = get_n_cntxt(n_cntxt)(X, Y)
+
+    model = copy.deepcopy(model).cpu().eval()
+    if length_scale_delta:
+        rbf = model.cntxt_to_induced.radial_basis_func
+        rbf.length_scale_param = _torch.nn.Parameter(
+            rbf.length_scale_param + length_scale_delta
+        )
+
+    X_plot = _torch.linspace(-1, 1, model.density_induced * 2).view(1, -1, 1)
+    with _torch.no_grad():
+        X_induced, R_setconv, R, R_trgt, p_yCc = _convcnp_stages(
+            model, X_cntxt, Y_cntxt, X_plot
+        )
+    burn = model.density_induced // 2
+    Xi = X_induced[0, burn:-burn, 0]
+
+    def stage_fig(curves, labels, title, show_cntxt=True, text=None):
+        with plot_config(**plot_config_kwargs):
+            fig, ax = plt.subplots(1, 1, figsize=(11, 5))
+            from npf.utils.helpers import rescale_range
+
+            for cur, lab in zip(curves, labels):
+                x, y, style = cur
+                ax.plot(
+                    rescale_range(x.numpy(), (-1, 1), dataset.min_max),
+                    y.numpy(), style, label=lab, alpha=0.8,
+                )
+            if show_cntxt:
+                ax.scatter(
+                    rescale_range(X_cntxt[0, :, 0].numpy(), (-1, 1), dataset.min_max),
+                    Y_cntxt[0, :, 0].numpy(), c="k", zorder=3,
+                )
+            if text is not None:
+                ax.text(
+                    0.5, 0.5, text, ha="center", va="center", fontsize=32,
+                    transform=ax.transAxes,
+                )
+            ax.set_xlim(list(dataset.min_max))
+            if title:
+                ax.set_title(title)
+            if labels and labels[0]:
+                ax.legend()
+        return fig
+
+    figs = []
+
+    def add(fig):
+        figs.append(fig2img(fig))
+        plt.close(fig)
+
+    # 1. the context set alone
+    add(stage_fig([], [], "Context set"))
+    # 2. announce the SetConv
+    add(stage_fig([], [], None, text="Apply SetConv"))
+    # 3. SetConv values at induced points
+    vals = R_setconv[0, burn:-burn, :-1].mean(-1)
+    add(stage_fig([(Xi, vals, "-")], ["SetConv"], "SetConv output"))
+    # 4. the density channel
+    dens = R_setconv[0, burn:-burn, -1]
+    add(stage_fig([(Xi, dens, "-")], ["density"], "Density channel"))
+    # 5. announce the CNN
+    add(stage_fig([], [], None, text="Apply CNN"))
+    # 6. a few post-CNN channels
+    chans = [(Xi, R[0, burn:-burn, c], "-") for c in range(3)]
+    add(stage_fig(chans, [f"channel {c}" for c in range(3)], "Post-CNN channels"))
+    # 7. the predictive
+    loc = p_yCc.base_dist.loc[0, 0, :, 0]
+    scale = p_yCc.base_dist.scale[0, 0, :, 0]
+    with plot_config(**plot_config_kwargs):
+        fig, ax = plt.subplots(1, 1, figsize=(11, 5))
+        from npf.utils.helpers import rescale_range
+
+        xs = rescale_range(X_plot[0, :, 0].numpy(), (-1, 1), dataset.min_max)
+        ax.plot(xs, loc.numpy(), "b-", label="Predictive mean")
+        ax.fill_between(
+            xs, (loc - scale).numpy(), (loc + scale).numpy(),
+            alpha=0.2, color="tab:blue",
+        )
+        ax.scatter(
+            rescale_range(X_cntxt[0, :, 0].numpy(), (-1, 1), dataset.min_max),
+            Y_cntxt[0, :, 0].numpy(), c="k", zorder=3,
+        )
+        ax.set_xlim(list(dataset.min_max))
+        ax.set_title("Predictive distribution")
+        ax.legend()
+    add(fig)
+
+    from PIL import Image
+
+    frames = [Image.fromarray(f).convert("P", palette=Image.ADAPTIVE) for f in figs]
+    w = min(f.size[0] for f in frames)
+    h = min(f.size[1] for f in frames)
+    frames = [f.resize((w, h)) for f in frames]
+    frames[0].save(
+        save_filename, save_all=True, append_images=frames[1:],
+        duration=int(1000 / fps), loop=0,
+    )
+    return save_filename
