@@ -13,6 +13,10 @@ REPO = os.path.abspath(os.path.join(os.path.dirname(__file__), ".."))
 
 
 def build():
+    # ALWAYS build clean: a stale ninja log (e.g. restored from a checkpoint
+    # without its object files) or a snapshot taken mid-rebuild can produce a
+    # .so that loads but memory-faults on the GPU — observed 2026-09-13.
+    shutil.rmtree(os.path.join(REPO, "csrc", "_build"), ignore_errors=True)
     os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
     os.environ.setdefault("MAX_JOBS", str(min(os.cpu_count() or 4, 16)))
 
