@@ -65,6 +65,13 @@ void npf_gde_fwd_launch(const float*, const float*, const float*, float*, int,
 void npf_gde_bwd_launch(const float*, const float*, const float*, const float*,
                         const float*, float*, float*, int, int, int, int, int,
                         hipStream_t);
+void npf_gauss_kl_fwd_launch(const float*, const float*, const float*,
+                             const float*, float*, long long, long long,
+                             hipStream_t);
+void npf_gauss_kl_bwd_launch(const float*, const float*, const float*,
+                             const float*, const float*, float*, float*,
+                             float*, float*, long long, long long,
+                             hipStream_t);
 }
 
 namespace {
@@ -401,6 +408,40 @@ std::tuple<torch::Tensor, torch::Tensor> griddensity_bwd(torch::Tensor x,
   return {dx, dw};
 }
 
+
+// fused diagonal-Gaussian KL + reduce: [B, ...] -> [B]
+torch::Tensor gauss_kl_fwd(torch::Tensor mq, torch::Tensor sq,
+                           torch::Tensor mp, torch::Tensor sp) {
+  for (auto* t : {&mq, &sq, &mp, &sp}) check_cuda_contig(*t, "kl input");
+  TORCH_CHECK(mq.scalar_type() == torch::kFloat32, "gauss_kl is fp32");
+  const long long rows = mq.size(0);
+  const long long m = mq.numel() / rows;
+  auto out = torch::empty({rows}, mq.options());
+  auto stream = at::hip::getCurrentHIPStream();
+  npf_gauss_kl_fwd_launch(mq.data_ptr<float>(), sq.data_ptr<float>(),
+                          mp.data_ptr<float>(), sp.data_ptr<float>(),
+                          out.data_ptr<float>(), rows, m, stream);
+  return out;
+}
+
+std::vector<torch::Tensor> gauss_kl_bwd(torch::Tensor mq, torch::Tensor sq,
+                                        torch::Tensor mp, torch::Tensor sp,
+                                        torch::Tensor dout) {
+  const long long rows = mq.size(0);
+  const long long m = mq.numel() / rows;
+  auto dmq = torch::empty_like(mq);
+  auto dsq = torch::empty_like(sq);
+  auto dmp = torch::empty_like(mp);
+  auto dsp = torch::empty_like(sp);
+  auto stream = at::hip::getCurrentHIPStream();
+  npf_gauss_kl_bwd_launch(mq.data_ptr<float>(), sq.data_ptr<float>(),
+                          mp.data_ptr<float>(), sp.data_ptr<float>(),
+                          dout.data_ptr<float>(), dmq.data_ptr<float>(),
+                          dsq.data_ptr<float>(), dmp.data_ptr<float>(),
+                          dsp.data_ptr<float>(), rows, m, stream);
+  return {dmq, dsq, dmp, dsp};
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -424,5 +465,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "fused masked abs-conv density encoder forward");
   m.def("griddensity_bwd", &griddensity_bwd,
         "fused density encoder backward -> (dx, dw)");
+  m.def("gauss_kl_fwd", &gauss_kl_fwd, "fused diagonal-Gaussian KL + reduce");
+  m.def("gauss_kl_bwd", &gauss_kl_bwd,
+        "fused KL backward -> (dmq, dsq, dmp, dsp)");
   m.attr("_arch") = "gfx950";
 }

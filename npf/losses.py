@@ -98,8 +98,20 @@ class ELBOLossLNPF(BaseLossNPF):
         # E_{q(z|C,T)}[ sum_t log p(y^t|z) ]
         E_z_sum_log_p_yCz = sum_log_prob(p_yCc, Y_trgt).mean(0)
         # sum_l KL[ q(z^l|C,T) || q(z^l|C) ]
-        kl_z = kl_divergence(q_zCct, q_zCc)
-        E_z_kl = sum_from_nth_dim(kl_z, 1)
+        if (
+            isinstance(q_zCct, Independent)
+            and isinstance(q_zCct.base_dist, Normal)
+            and isinstance(q_zCc, Independent)
+            and isinstance(q_zCc.base_dist, Normal)
+        ):
+            # fused HIP kernel: analytic KL + latent-set reduce in one pass
+            E_z_kl = ops.gaussian_kl_sum(
+                q_zCct.base_dist.loc, q_zCct.base_dist.scale,
+                q_zCc.base_dist.loc, q_zCc.base_dist.scale,
+            )
+        else:
+            kl_z = kl_divergence(q_zCct, q_zCc)
+            E_z_kl = sum_from_nth_dim(kl_z, 1)
         return -(E_z_sum_log_p_yCz - E_z_kl)
 
 

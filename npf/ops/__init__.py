@@ -2,13 +2,15 @@
 
 from ._backend import extension, has_extension, require_extension
 from .functional import (attention_qkv, conv_block_1d, conv_block_2d,
-                         gaussian_nll_sum, grid_density, setconv_gaussian)
+                         gaussian_kl_sum, gaussian_nll_sum, grid_density,
+                         setconv_gaussian)
 
 __all__ = [
     "attention_qkv",
     "conv_block_1d",
     "conv_block_2d",
     "grid_density",
+    "gaussian_kl_sum",
     "setconv_gaussian",
     "gaussian_nll_sum",
     "extension",

@@ -12,7 +12,7 @@ import torch
 
 from . import _backend
 
-__all__ = ["attention_qkv", "setconv_gaussian", "gaussian_nll_sum", "conv_block_1d", "conv_block_2d", "grid_density"]
+__all__ = ["attention_qkv", "setconv_gaussian", "gaussian_nll_sum", "conv_block_1d", "conv_block_2d", "grid_density", "gaussian_kl_sum"]
 
 
 # --------------------------------------------------------------------------- #
@@ -530,4 +530,45 @@ def grid_density(x, mask, weight):
     mask = mask.float().expand_as(x).contiguous()
     return _GridDensityFn.apply(
         x.float().contiguous(), mask, weight.float()
+    )
+
+
+# --------------------------------------------------------------------------- #
+# Fused diagonal-Gaussian KL + reduce (NPVI ELBO term).
+# Reference computation: losses.py:135-150 (kl_divergence(Normal, Normal)
+# composed with sum over latent dims).
+# --------------------------------------------------------------------------- #
+
+
+def _kl_ref(mq, sq, mp, sp):
+    kl = (
+        (sp / sq).log().neg().neg()  # log(sp) - log(sq) spelled for clarity
+        + (sq ** 2 + (mq - mp) ** 2) / (2 * sp ** 2)
+        - 0.5
+    )
+    return kl.reshape(kl.shape[0], -1).sum(-1)
+
+
+class _GaussKLFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, mq, sq, mp, sp):
+        ext = _backend.require_extension("gaussian_kl_sum")
+        ctx.save_for_backward(mq, sq, mp, sp)
+        return ext.gauss_kl_fwd(mq, sq, mp, sp)
+
+    @staticmethod
+    def backward(ctx, dout):
+        ext = _backend.require_extension("gaussian_kl_sum")
+        mq, sq, mp, sp = ctx.saved_tensors
+        dmq, dsq, dmp, dsp = ext.gauss_kl_bwd(mq, sq, mp, sp, dout.contiguous())
+        return dmq, dsq, dmp, dsp
+
+
+def gaussian_kl_sum(mq, sq, mp, sp):
+    """KL( N(mq, sq) || N(mp, sp) ) summed over all dims but the first."""
+    if not mq.is_cuda or _backend.require_extension("gaussian_kl_sum") is None:
+        return _kl_ref(mq, sq, mp, sp)
+    return _GaussKLFn.apply(
+        mq.float().contiguous(), sq.float().contiguous(),
+        mp.float().contiguous(), sp.float().contiguous(),
     )

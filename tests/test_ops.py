@@ -389,3 +389,24 @@ class TestHIPKernels:
         ref = m0(mc, Y, mt)
         a, b = out[0].base_dist.loc.cpu(), ref[0].base_dist.loc
         assert torch.allclose(a, b, atol=5e-3), (a - b).abs().max()
+
+    @pytest.mark.parametrize("B,M,Z", [(4, 7, 3), (32, 1, 128), (16, 192, 16)])
+    def test_gaussian_kl_fwd_bwd(self, B, M, Z):
+        from npf.ops.functional import _kl_ref
+
+        g = torch.Generator(device="cuda").manual_seed(0)
+        mk = lambda: torch.randn(B, M, Z, device="cuda", generator=g, requires_grad=True)
+        sk = lambda: (torch.rand(B, M, Z, device="cuda", generator=g) + 0.1
+                      ).requires_grad_(True)
+        mq, sq, mp, sp = mk(), sk(), mk(), sk()
+        cpu = [t.detach().cpu().requires_grad_(True) for t in (mq, sq, mp, sp)]
+        out = F_ops.gaussian_kl_sum(mq, sq, mp, sp)
+        ref = _kl_ref(*cpu)
+        assert torch.allclose(out.cpu(), ref, atol=1e-3), (out.cpu() - ref).abs().max()
+        dout = torch.randn_like(out)
+        out.backward(dout)
+        ref.backward(dout.cpu())
+        for a, b in zip((mq, sq, mp, sp), cpu):
+            assert torch.allclose(a.grad.cpu(), b.grad, atol=1e-3), (
+                (a.grad.cpu() - b.grad).abs().max()
+            )
