@@ -29,10 +29,10 @@ void npf_attn_bwd_launch_bf16(const void*, const void*, const void*,
                               void*, void*, int, int, int, int, int, float,
                               hipStream_t);
 void npf_setconv_fwd_launch(const float*, const float*, const float*, float*,
-                            int, int, int, int, float, hipStream_t);
+                            int, int, int, int, const float*, hipStream_t);
 void npf_setconv_bwd_launch(const float*, const float*, const float*,
                             const float*, float*, float*, float*, float*, int,
-                            int, int, int, float, hipStream_t);
+                            int, int, int, const float*, hipStream_t);
 void npf_gauss_ll_fwd_launch(const float*, const float*, const float*, float*,
                              long long, long long, hipStream_t);
 void npf_gauss_ll_bwd_launch(const float*, const float*, const float*,
@@ -145,7 +145,9 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> attn_bwd(
 }
 
 torch::Tensor setconv_fwd(torch::Tensor keys, torch::Tensor queries,
-                          torch::Tensor values, double sigma) {
+                          torch::Tensor values, torch::Tensor sigma) {
+  check_cuda_contig(sigma, "sigma");
+  TORCH_CHECK(sigma.scalar_type() == torch::kFloat32, "sigma must be fp32");
   check_cuda_contig(keys, "keys");
   check_cuda_contig(queries, "queries");
   check_cuda_contig(values, "values");
@@ -157,13 +159,13 @@ torch::Tensor setconv_fwd(torch::Tensor keys, torch::Tensor queries,
   auto stream = at::hip::getCurrentHIPStream();
   npf_setconv_fwd_launch(keys.data_ptr<float>(), queries.data_ptr<float>(),
                          values.data_ptr<float>(), out.data_ptr<float>(), B, K,
-                         Q, C, (float)sigma, stream);
+                         Q, C, sigma.data_ptr<float>(), stream);
   return out;
 }
 
 std::tuple<torch::Tensor, torch::Tensor, torch::Tensor, torch::Tensor>
 setconv_bwd(torch::Tensor keys, torch::Tensor queries, torch::Tensor values,
-            double sigma, torch::Tensor dout) {
+            torch::Tensor sigma, torch::Tensor dout) {
   check_cuda_contig(keys, "keys");
   check_cuda_contig(queries, "queries");
   check_cuda_contig(values, "values");
@@ -179,7 +181,7 @@ setconv_bwd(torch::Tensor keys, torch::Tensor queries, torch::Tensor values,
                          values.data_ptr<float>(), dout.data_ptr<float>(),
                          dk.data_ptr<float>(), dq.data_ptr<float>(),
                          dv.data_ptr<float>(), dsigma.data_ptr<float>(), B, K,
-                         Q, C, (float)sigma, stream);
+                         Q, C, sigma.data_ptr<float>(), stream);
   return {dk, dq, dv, dsigma};
 }
 

@@ -38,7 +38,9 @@
 extern "C" __global__ void __launch_bounds__(SC_BLOCK)
 npf_setconv_fwd(const float* __restrict__ xk, const float* __restrict__ xq,
                 const float* __restrict__ v, float* __restrict__ out,
-                int B, int K, int Q, int C, float sigma) {
+                int B, int K, int Q, int C,
+                const float* __restrict__ sigma_ptr) {
+  const float sigma = *sigma_ptr;
   extern __shared__ float smem[];
   float* w = smem;                       // [SC_QT][K]
   float* mq = smem + SC_QT * K;          // [SC_QT]
@@ -96,7 +98,9 @@ npf_setconv_bwd(const float* __restrict__ xk, const float* __restrict__ xq,
                 const float* __restrict__ v, const float* __restrict__ dout,
                 float* __restrict__ dxk, float* __restrict__ dxq,
                 float* __restrict__ dv, float* __restrict__ dsigma,
-                int B, int K, int Q, int C, float sigma) {
+                int B, int K, int Q, int C,
+                const float* __restrict__ sigma_ptr) {
+  const float sigma = *sigma_ptr;
   extern __shared__ float smem[];
   float* w = smem;                        // [SC_QT][K] normalized weights
   float* g = smem + SC_QT * K;            // [SC_QT][K] dots -> dL/ds
@@ -213,7 +217,7 @@ npf_setconv_bwd(const float* __restrict__ xk, const float* __restrict__ xq,
 
 extern "C" void npf_setconv_fwd_launch(const float* xk, const float* xq,
                                        const float* v, float* out, int B,
-                                       int K, int Q, int C, float sigma,
+                                       int K, int Q, int C, const float* sigma,
                                        hipStream_t stream) {
   dim3 grid(B, (Q + SC_QT - 1) / SC_QT);
   const size_t smem = (size_t)(SC_QT * K + 2 * SC_QT) * sizeof(float);
@@ -225,7 +229,7 @@ extern "C" void npf_setconv_bwd_launch(const float* xk, const float* xq,
                                        const float* v, const float* dout,
                                        float* dxk, float* dxq, float* dv,
                                        float* dsigma, int B, int K, int Q,
-                                       int C, float sigma,
+                                       int C, const float* sigma,
                                        hipStream_t stream) {
   dim3 grid(B, (Q + SC_QT - 1) / SC_QT);
   const size_t smem =

@@ -129,18 +129,24 @@ class _SetConvFn(torch.autograd.Function):
             ctx.save_for_backward(keys, queries, values, sigma)
             ctx.fused = False
             return out
-        out = ext.setconv_fwd(keys, queries, values, float(sigma))
-        ctx.save_for_backward(keys, queries, values, sigma)
+        # sigma rides as a device scalar: no host sync (hipGraph-capture
+        # safe) and graph replays see the CURRENT learned length-scale
+        sig32 = sigma.detach().reshape(1).float().contiguous()
+        out = ext.setconv_fwd(keys, queries, values, sig32)
+        ctx.save_for_backward(keys, queries, values, sigma, sig32)
         ctx.fused = True
         return out
 
     @staticmethod
     def backward(ctx, dout):
-        keys, queries, values, sigma = ctx.saved_tensors
+        if ctx.fused:
+            keys, queries, values, sigma, sig32 = ctx.saved_tensors
+        else:
+            keys, queries, values, sigma = ctx.saved_tensors
         ext = _backend.extension()
         if ctx.fused and ext is not None:
             dout = dout.contiguous()
-            dk, dq, dv, dsig = ext.setconv_bwd(keys, queries, values, float(sigma), dout)
+            dk, dq, dv, dsig = ext.setconv_bwd(keys, queries, values, sig32, dout)
             return dk, dq, dv, dsig.to(sigma.dtype).reshape(sigma.shape)
 
         # composed fallback (double-backward capable is not required)
@@ -393,7 +399,7 @@ class _ConvBlock2dFn(torch.autograd.Function):
         ext = _backend.require_extension("conv_block_2d")
         C = x.shape[1]
         K = weight.shape[-1]
-        w2d = weight.view(C, K * K).contiguous()
+        w2d = weight.view(C, K, K).contiguous()
         has_bn = gamma is not None
         und = torch.Tensor()
         if has_bn:
@@ -507,7 +513,7 @@ class _GridDensityFn(torch.autograd.Function):
         ext = _backend.require_extension("grid_density")
         C = x.shape[1]
         K = weight.shape[-1]
-        w2d = weight.view(C, K * K).contiguous()
+        w2d = weight.view(C, K, K).contiguous()
         out = ext.griddensity_fwd(x, mask, w2d)
         ctx.save_for_backward(x, mask, w2d, out)
         ctx.kshape = weight.shape
