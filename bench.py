@@ -216,9 +216,11 @@ def main():
         assert world == args.gpus or args.gpus == 1, (world, args.gpus)
     n_gpus = world if world > 1 else 1
 
-    if use_cuda:
+    if use_cuda and not os.environ.get("NPF_BENCH_NO_TUNE"):
         # per-shape GEMM autotuning (rocBLAS/hipBLASLt/CK): tunes during
-        # warmup, then the tuned kernels are what the graph captures
+        # warmup, then the tuned kernels are what the graph captures.
+        # NPF_BENCH_NO_TUNE=1 skips it (profiling runs: tuning floods the
+        # kernel stats with thousands of one-off candidate launches)
         try:
             torch.cuda.tunable.enable(True)
             torch.cuda.tunable.tuning_enable(True)
