@@ -175,6 +175,10 @@ def _configs():
             batch=32,
             pool=lambda dev, b, s: _img_point_pool(dev, b, s, shape=(3, 32, 32)),
             needs_y_trgt=True,  # NPVI: q_zCct from the target set
+            graph_ok=False,  # latent rsample inside a hipGraph replays
+                             # corrupted noise on ROCm -> loss NaN; the step
+                             # is compute-bound (~11 ms) so replay gains
+                             # nothing anyway
             desc="AttnLNP-2D CelebA32-shape (self-attn encoder, NPVI, 468,486 params)",
             seq_len=32 * 32,
             n_cntxt=int(0.3 * 32 * 32),
@@ -185,6 +189,7 @@ def _configs():
             batch=16,
             pool=lambda dev, b, s: _img_grid_pool(dev, b, s, shape=(3, 64, 64)),
             needs_y_trgt=False,  # NPML
+            graph_ok=False,  # see attnlnp2d
             desc="GridConvLNP-2D CelebA64-shape (4+4-block CNN k=9, NPML z=16, 487,793 params)",
             seq_len=64 * 64,
             n_cntxt=int(0.1 * 64 * 64),
@@ -293,7 +298,7 @@ def main():
 
     # ---- optional hipGraph capture of the whole train step ----
     graph = None
-    if use_cuda and not args.no_graph:
+    if use_cuda and not args.no_graph and cfg.get("graph_ok", True):
         import gc
 
         # drop every reference to the warmup autograd graph: a live
