@@ -116,6 +116,10 @@ npf_cb_fwd(const float* __restrict__ x, const float* __restrict__ res,
 
 // B1: dact = relu'(a) * corr(w, dY); per-channel partials for BN/weight grads
 //     dact is stored; channel sums go through atomics on [C] buffers.
+//     One workgroup covers CB_TN batch rows of ONE channel so the K
+//     block-reductions for dW amortize over the tile (measured 72us -> the
+//     per-(n,c) version spent most of its time in 19 serialized reductions).
+#define CB_TN 4
 extern "C" __global__ void __launch_bounds__(CB_BLOCK)
 npf_cb_bwd_dact(const float* __restrict__ x, const float* __restrict__ w,
                 const float* __restrict__ dy, const float* __restrict__ gamma,
@@ -126,29 +130,29 @@ npf_cb_bwd_dact(const float* __restrict__ x, const float* __restrict__ w,
                 float* __restrict__ dgamma, float* __restrict__ dbeta,
                 int N, int C, int L, int K) {
   extern __shared__ float smem[];
-  float* dyrow_s = smem;            // [L + K - 1] dY row with halo
-  float* arow_s = smem + L + K - 1; // [L + K - 1] activation row with halo
+  const int pad = K / 2;
+  const int LP = L + 2 * pad;
+  float* dys = smem;            // [CB_TN][LP] dY rows with halo
+  float* as = smem + CB_TN * LP;  // [CB_TN][LP] activation rows with halo
   __shared__ float red[16];
 
-  const int n = blockIdx.x / C;
   const int c = blockIdx.x % C;
-  const int pad = K / 2;
-  const float* xrow = x + ((long)n * C + c) * L;
-  const float* dyrow = dy + ((long)n * C + c) * L;
+  const int n0 = (blockIdx.x / C) * CB_TN;
+  const int tn = min(CB_TN, N - n0);
 
   const bool has_bn = gamma != nullptr;
   const float mu = has_bn ? mean[c] : 0.f;
   const float rs = has_bn ? rstd[c] : 1.f;
-  const float gscale = has_bn ? rs * gamma[c] : 1.f;
+  const float gm = has_bn ? gamma[c] : 1.f;
+  const float gscale = has_bn ? rs * gm : 1.f;
   const float gshift = has_bn ? beta[c] : 0.f;
 
-  for (int l = threadIdx.x; l < L + 2 * pad; l += CB_BLOCK) {
-    const int src = l - pad;
-    dyrow_s[l] = (src >= 0 && src < L) ? dyrow[src] : 0.f;
-    float v = 0.f;
-    if (src >= 0 && src < L)
-      v = fmaxf((xrow[src] - mu) * gscale + gshift, 0.f);
-    arow_s[l] = v;
+  for (int i = threadIdx.x; i < tn * LP; i += CB_BLOCK) {
+    const int t = i / LP, src = i % LP - pad;
+    const long base = ((long)(n0 + t) * C + c) * L;
+    const bool in = (src >= 0 && src < L);
+    dys[i] = in ? dy[base + src] : 0.f;
+    as[i] = in ? fmaxf((x[base + src] - mu) * gscale + gshift, 0.f) : 0.f;
   }
   __syncthreads();
 
@@ -156,16 +160,18 @@ npf_cb_bwd_dact(const float* __restrict__ x, const float* __restrict__ w,
   #pragma unroll 4
   for (int j = 0; j < K; ++j) wreg[j] = w[c * K + j];
 
-  // dW[c,j] = sum_l dY[l] * a[l+j-pad]; db[c] = sum_l dY[l]
+  // dW[c,j] = sum_{n,l} dY * a[l+j-pad]; db[c] = sum dY
   float dwp[CB_MAX_K];
   #pragma unroll 4
   for (int j = 0; j < K; ++j) dwp[j] = 0.f;
   float dbp = 0.f;
 
-  float* dactrow = dact + ((long)n * C + c) * L;
   float s_dxhat = 0.f, s_dxhat_xhat = 0.f, s_dg = 0.f, s_db = 0.f;
-  for (int l = threadIdx.x; l < L; l += CB_BLOCK) {
-    const float dyl = dyrow[l];
+  for (int i = threadIdx.x; i < tn * L; i += CB_BLOCK) {
+    const int t = i / L, l = i % L;
+    const float* dyrow_s = dys + t * LP;
+    const float* arow_s = as + t * LP;
+    const float dyl = dyrow_s[l + pad];
     dbp += dyl;
     #pragma unroll 4
     for (int j = 0; j < K; ++j) dwp[j] += dyl * arow_s[l + j];
@@ -175,10 +181,10 @@ npf_cb_bwd_dact(const float* __restrict__ x, const float* __restrict__ w,
     for (int j = 0; j < K; ++j) da += wreg[j] * dyrow_s[l + (K - 1 - j)];
     const float act = arow_s[l + pad];
     const float dr = (act > 0.f) ? da : 0.f;  // through relu
-    dactrow[l] = dr;
+    dact[((long)(n0 + t) * C + c) * L + l] = dr;
     if (has_bn) {
-      const float xhat = (xrow[l] - mu) * rs;
-      const float dxh = dr * gamma[c];
+      const float xhat = (x[((long)(n0 + t) * C + c) * L + l] - mu) * rs;
+      const float dxh = dr * gm;
       s_dxhat += dxh;
       s_dxhat_xhat += dxh * xhat;
       s_dg += dr * xhat;
@@ -283,8 +289,9 @@ extern "C" void npf_cb_bwd_launch(const float* x, const float* w,
                                   float* dbeta, float* dx, int N, int C,
                                   int L, int K, int training,
                                   hipStream_t stream) {
-  const size_t smem = (size_t)(2 * (L + K - 1)) * sizeof(float);
-  hipLaunchKernelGGL(npf_cb_bwd_dact, dim3((unsigned)N * C), dim3(CB_BLOCK),
+  const size_t smem = (size_t)(2 * CB_TN * (L + K - 1)) * sizeof(float);
+  const unsigned ntiles = (unsigned)((N + CB_TN - 1) / CB_TN);
+  hipLaunchKernelGGL(npf_cb_bwd_dact, dim3(ntiles * C), dim3(CB_BLOCK),
                      smem, stream, x, w, dy, gamma, beta, mean, rstd, dact,
                      dw, db, sum_dxhat, sum_dxhat_xhat, dgamma, dbeta, N, C,
                      L, K);
