@@ -16,8 +16,9 @@
 // backward in 3, keeping the activation row staged in LDS with its conv
 // halo.
 //
-// Launch shape: apply kernels use one workgroup per (n, c) row, 256 threads
-// striding the L dimension -> N*C = 4k-65k workgroups (fills 256 CUs);
+// Launch shape: the forward apply kernel uses one workgroup per (n, c) row
+// (N*C = 4k-65k workgroups fills 256 CUs); the backward stencil tiles CB_TN
+// batch rows per workgroup so its per-weight block-reductions amortize;
 // stats/reduction kernels use one workgroup per channel.
 //
 // fp32 compute (BN statistics need it; tensors are fp32 master or bf16 —
