@@ -31,6 +31,7 @@ def build():
         os.path.join(REPO, "csrc", "npf_hip", "convblock2d.hip"),
         os.path.join(REPO, "csrc", "npf_hip", "griddensity.hip"),
         os.path.join(REPO, "csrc", "npf_hip", "gauss_kl.hip"),
+        os.path.join(REPO, "csrc", "npf_hip", "mlp_chain.hip"),
     ]
     ext = CUDAExtension(
         name="npf._hip_C",

@@ -72,6 +72,12 @@ void npf_gauss_kl_bwd_launch(const float*, const float*, const float*,
                              const float*, const float*, float*, float*,
                              float*, float*, long long, long long,
                              hipStream_t);
+void npf_mlp_fwd_launch(const void*, const float* const*, const float* const*,
+                        void* const*, void*, const int*, int, long,
+                        hipStream_t);
+void npf_mlp_bwd_launch(const void*, const float* const*, const void* const*,
+                        void* const*, void*, float* const*, const int*, int,
+                        long, hipStream_t);
 }
 
 namespace {
@@ -444,6 +450,85 @@ std::vector<torch::Tensor> gauss_kl_bwd(torch::Tensor mq, torch::Tensor sq,
   return {dmq, dsq, dmp, dsp};
 }
 
+
+// fused MFMA MLP chain: x [R, d0] bf16 -> y [R, dL] bf16 (+ saved acts)
+std::vector<torch::Tensor> mlp_chain_fwd(torch::Tensor x,
+                                         std::vector<torch::Tensor> ws,
+                                         std::vector<torch::Tensor> bs) {
+  check_cuda_contig(x, "x");
+  TORCH_CHECK(x.scalar_type() == torch::kBFloat16, "mlp_chain x must be bf16");
+  const int L = (int)ws.size();
+  TORCH_CHECK(L >= 2 && L <= 8, "mlp_chain supports 2..8 layers");
+  const long R = x.numel() / x.size(-1);
+  int d[9];
+  d[0] = (int)x.size(-1);
+  const float* wp[8];
+  const float* bp[8];
+  void* ap[8] = {nullptr};
+  std::vector<torch::Tensor> acts;
+  for (int i = 0; i < L; ++i) {
+    check_cuda_contig(ws[i], "w");
+    TORCH_CHECK(ws[i].scalar_type() == torch::kFloat32, "weights must be fp32");
+    TORCH_CHECK(ws[i].size(1) == d[i], "layer dim mismatch");
+    d[i + 1] = (int)ws[i].size(0);
+    TORCH_CHECK(d[i + 1] <= 128 && d[i] <= 128, "mlp_chain dims <= 128");
+    wp[i] = ws[i].data_ptr<float>();
+    bp[i] = bs[i].data_ptr<float>();
+    if (i < L - 1) {
+      acts.push_back(torch::empty({R, (long)d[i + 1]},
+                                  x.options().dtype(torch::kBFloat16)));
+      ap[i] = acts.back().data_ptr();
+    }
+  }
+  auto y = torch::empty({R, (long)d[L]}, x.options().dtype(torch::kBFloat16));
+  auto stream = at::hip::getCurrentHIPStream();
+  npf_mlp_fwd_launch(x.data_ptr(), wp, bp, ap, y.data_ptr(), d, L, R, stream);
+  std::vector<torch::Tensor> out = {y};
+  for (auto& a : acts) out.push_back(a);
+  return out;
+}
+
+// backward: returns {dx_or_empty, dz_0..dz_{L-1}, db_0..db_{L-1}}
+std::vector<torch::Tensor> mlp_chain_bwd(torch::Tensor dy,
+                                         std::vector<torch::Tensor> ws,
+                                         std::vector<torch::Tensor> acts,
+                                         bool need_dx) {
+  check_cuda_contig(dy, "dy");
+  TORCH_CHECK(dy.scalar_type() == torch::kBFloat16, "dy must be bf16");
+  const int L = (int)ws.size();
+  const long R = dy.numel() / dy.size(-1);
+  int d[9];
+  const float* wp[8];
+  const void* ap[8] = {nullptr};
+  void* zp[8];
+  float* dbp[8];
+  for (int i = 0; i < L; ++i) {
+    wp[i] = ws[i].data_ptr<float>();
+    d[i] = (int)ws[i].size(1);
+    d[i + 1] = (int)ws[i].size(0);
+    if (i < L - 1) ap[i] = acts[i].data_ptr();
+  }
+  std::vector<torch::Tensor> dzs, dbs;
+  for (int i = 0; i < L; ++i) {
+    dzs.push_back(torch::empty({R, (long)d[i + 1]},
+                               dy.options().dtype(torch::kBFloat16)));
+    zp[i] = dzs.back().data_ptr();
+    dbs.push_back(torch::zeros({(long)d[i + 1]},
+                               dy.options().dtype(torch::kFloat32)));
+    dbp[i] = dbs.back().data_ptr<float>();
+  }
+  torch::Tensor dx;
+  if (need_dx)
+    dx = torch::empty({R, (long)d[0]}, dy.options().dtype(torch::kBFloat16));
+  auto stream = at::hip::getCurrentHIPStream();
+  npf_mlp_bwd_launch(dy.data_ptr(), wp, ap, zp,
+                     need_dx ? dx.data_ptr() : nullptr, dbp, d, L, R, stream);
+  std::vector<torch::Tensor> out = {need_dx ? dx : torch::Tensor()};
+  for (auto& z : dzs) out.push_back(z);
+  for (auto& b : dbs) out.push_back(b);
+  return out;
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -470,5 +555,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gauss_kl_fwd", &gauss_kl_fwd, "fused diagonal-Gaussian KL + reduce");
   m.def("gauss_kl_bwd", &gauss_kl_bwd,
         "fused KL backward -> (dmq, dsq, dmp, dsp)");
+  m.def("mlp_chain_fwd", &mlp_chain_fwd,
+        "fused MFMA MLP chain forward -> [y, act_0..act_{L-2}]");
+  m.def("mlp_chain_bwd", &mlp_chain_bwd,
+        "fused MLP chain backward -> [dx, dz_0.., db_0..]");
   m.attr("_arch") = "gfx950";
 }

@@ -9,6 +9,7 @@ fusion candidates for the HIP epilogue kernels (SURVEY.md §2.3 "MLP decoder").
 
 import warnings
 
+import torch
 import torch.nn as nn
 
 from npf.utils.initialization import linear_init
@@ -69,7 +70,35 @@ class MLP(nn.Module):
 
         self.reset_parameters()
 
+    def _fused_ok(self, x):
+        from npf.ops import has_extension
+
+        return (
+            x.is_cuda
+            and torch.is_autocast_enabled("cuda")
+            and isinstance(self.activation, nn.ReLU)
+            and isinstance(self.dropout, nn.Identity)
+            and not self.is_res
+            and self.input_size <= 128
+            and self.hidden_size <= 128
+            and self.output_size <= 128
+            and self.to_hidden.bias is not None
+            and self.to_hidden.weight.dtype == torch.float32
+            and has_extension()
+        )
+
     def forward(self, x):
+        if self._fused_ok(x):
+            # whole chain in ONE MFMA kernel (csrc/npf_hip/mlp_chain.hip):
+            # bf16 compute + fp32 accumulate = the autocast numerics this
+            # branch replaces, minus ~3 kernels per layer
+            from npf.ops import mlp_chain
+
+            layers = [self.to_hidden, *self.linears, self.out]
+            return mlp_chain(
+                x, [l.weight for l in layers], [l.bias for l in layers]
+            )
+
         # first layer: linear -> act -> dropout (reference mlp.py:95-98)
         h = self.dropout(self.activation(self.to_hidden(x)))
         for linear in self.linears:

@@ -3,7 +3,7 @@
 from ._backend import extension, has_extension, require_extension
 from .functional import (attention_qkv, conv_block_1d, conv_block_2d,
                          gaussian_kl_sum, gaussian_nll_sum, grid_density,
-                         setconv_gaussian)
+                         mlp_chain, setconv_gaussian)
 
 __all__ = [
     "attention_qkv",
@@ -11,6 +11,7 @@ __all__ = [
     "conv_block_2d",
     "grid_density",
     "gaussian_kl_sum",
+    "mlp_chain",
     "setconv_gaussian",
     "gaussian_nll_sum",
     "extension",

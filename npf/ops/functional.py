@@ -12,7 +12,7 @@ import torch
 
 from . import _backend
 
-__all__ = ["attention_qkv", "setconv_gaussian", "gaussian_nll_sum", "conv_block_1d", "conv_block_2d", "grid_density", "gaussian_kl_sum"]
+__all__ = ["attention_qkv", "setconv_gaussian", "gaussian_nll_sum", "conv_block_1d", "conv_block_2d", "grid_density", "gaussian_kl_sum", "mlp_chain"]
 
 
 # --------------------------------------------------------------------------- #
@@ -578,3 +578,56 @@ def gaussian_kl_sum(mq, sq, mp, sp):
         mq.float().contiguous(), sq.float().contiguous(),
         mp.float().contiguous(), sp.float().contiguous(),
     )
+
+
+# --------------------------------------------------------------------------- #
+# Fused MFMA MLP chain (csrc/npf_hip/mlp_chain.hip).
+# Reference computation: npf/architectures/mlp.py MLP.forward — a chain of
+# torch Linears with ReLU between them (reference mlp.py:95-109).
+# --------------------------------------------------------------------------- #
+
+
+class _MLPChainFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, n_layers, *wb):
+        ext = _backend.require_extension("mlp_chain")
+        ws = list(wb[:n_layers])
+        bs = list(wb[n_layers:])
+        out = ext.mlp_chain_fwd(x, ws, bs)
+        y, acts = out[0], out[1:]
+        ctx.save_for_backward(x, *ws, *acts)
+        ctx.n_layers = n_layers
+        ctx.need_dx = x.requires_grad
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        ext = _backend.require_extension("mlp_chain")
+        L = ctx.n_layers
+        x = ctx.saved_tensors[0]
+        ws = list(ctx.saved_tensors[1 : 1 + L])
+        acts = list(ctx.saved_tensors[1 + L :])
+        out = ext.mlp_chain_bwd(dy.to(torch.bfloat16).contiguous(), ws, acts,
+                                ctx.need_dx)
+        dx = out[0] if ctx.need_dx else None
+        dzs = out[1 : 1 + L]
+        dbs = out[1 + L :]
+        # dW_l = dz_l^T @ a_{l-1}: K-large library GEMMs, fp32 master grads
+        grads = [dx, None]
+        ins = [x] + acts
+        dws = [
+            torch.mm(dzs[l].t(), ins[l]).float() for l in range(L)
+        ]
+        return tuple(grads + dws + list(dbs))
+
+
+def mlp_chain(x, weights, biases):
+    """Fused MLP chain: y = W_L(relu(... relu(W_1 x + b_1) ...)) + b_L.
+
+    x [..., d0] (any float dtype; computed in bf16 with fp32 accumulation =
+    autocast semantics), weights fp32 [out, in] per layer, out bf16.
+    """
+    lead = x.shape[:-1]
+    xb = x.reshape(-1, x.shape[-1]).to(torch.bfloat16).contiguous()
+    y = _MLPChainFn.apply(xb, len(weights), *weights, *biases)
+    return y.reshape(*lead, y.shape[-1])
