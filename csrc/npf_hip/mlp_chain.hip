@@ -25,6 +25,8 @@
 //   A: lane holds row = lane&15, k = (lane>>4)*8 .. +8   (8 bf16 = 4 VGPRs)
 //   B: lane holds col = lane&15, k = (lane>>4)*8 .. +8
 //   C/D: lane holds col = lane&15, row = (lane>>4)*4 + reg (4 fp32)
+// Padded dims round up to 32: the MFMA consumes a full K=32 slice per
+// issue, so staging must zero-fill the whole 32-wide contraction chunk.
 // For y = x @ W^T with torch Linear W[out,in], the B fragment (col=out,
 // k=in) reads contiguous rows of W (no transpose); the backward's
 // da = dz @ W stages W transposed.
@@ -107,7 +109,7 @@ npf_mlp_fwd(MlpParams p) {
   const int wave = threadIdx.x >> 6;
 
   const int d0 = p.d[0];
-  const int d0p = (d0 + 15) & ~15;
+  const int d0p = (d0 + 31) & ~31;
   for (int i = threadIdx.x; i < MC_TR * d0p; i += MC_BLOCK) {
     const int r = i / d0p, c = i % d0p;
     a_lds[0][r][c] = (r0 + r < p.R && c < d0)
@@ -118,7 +120,7 @@ npf_mlp_fwd(MlpParams p) {
   int cur = 0;
   for (int l = 0; l < p.L; ++l) {
     const int din = p.d[l], dout = p.d[l + 1];
-    const int din_p = (din + 15) & ~15, dout_p = (dout + 15) & ~15;
+    const int din_p = (din + 31) & ~31, dout_p = (dout + 31) & ~31;
     stage_w(p.w[l], w_lds, dout, din, dout_p, din_p, false);
     for (int i = threadIdx.x; i < dout_p; i += MC_BLOCK)
       b_lds[i] = (i < dout) ? p.b[l][i] : 0.f;
@@ -184,7 +186,7 @@ npf_mlp_bwd(MlpParams p) {
 
   // stage dz_{L-1} = dY; store it and its db
   const int dl = p.d[L];
-  const int dlp = (dl + 15) & ~15;
+  const int dlp = (dl + 31) & ~31;
   for (int i = threadIdx.x; i < MC_TR * dlp; i += MC_BLOCK) {
     const int r = i / dlp, c = i % dlp;
     a_lds[0][r][c] = (r0 + r < p.R && c < dl)
@@ -201,7 +203,7 @@ npf_mlp_bwd(MlpParams p) {
   int cur = 0;
   for (int l = L - 1; l >= 1; --l) {
     const int din = p.d[l], dout = p.d[l + 1];
-    const int din_p = (din + 15) & ~15, dout_p = (dout + 15) & ~15;
+    const int din_p = (din + 31) & ~31, dout_p = (dout + 31) & ~31;
     // da_{l-1} = dz_l @ W_l : stage W transposed so B[k=out][n=in]
     stage_w(p.w[l], w_lds, dout, din, dout_p, din_p, true);
     __syncthreads();
@@ -241,7 +243,7 @@ npf_mlp_bwd(MlpParams p) {
   // dX = dz_0 @ W_0
   if (p.dx != nullptr) {
     const int din = p.d[0], dout = p.d[1];
-    const int din_p = (din + 15) & ~15, dout_p = (dout + 15) & ~15;
+    const int din_p = (din + 31) & ~31, dout_p = (dout + 31) & ~31;
     stage_w(p.w[0], w_lds, dout, din, dout_p, din_p, true);
     __syncthreads();
     const int row0 = wave * 16;
