@@ -437,9 +437,10 @@ class TestHIPKernels:
 
         out = F_ops.mlp_chain(x, ws, bs)
 
-        # bf16-eager oracle (same rounding points: bf16 GEMM in, bf16 out);
+        # bf16-eager oracle (same rounding points: bf16 GEMM in, bf16 out,
+        # weights rounded to bf16 exactly like the kernel's LDS staging);
         # detached weight copies so its backward doesn't pollute ws/bs grads
-        wd = [w.detach() for w in ws]
+        wd = [w.detach().to(torch.bfloat16).float() for w in ws]
         bd = [b.detach() for b in bs]
 
         def ref(xb):
