@@ -65,18 +65,88 @@ __device__ __forceinline__ bf16x8 ld_frag16(const __hip_bfloat16* p) {
   return *reinterpret_cast<const bf16x8*>(p);  // 16B-aligned by layout
 }
 
+
+// stage a [MC_TR, d] bf16 global tile (rows r0..r0+MC_TR of [R, d]) into
+// LDS, zero-padding cols to d_p and rows beyond R; 16B loads when d % 8 == 0
+__device__ __forceinline__ void stage_x(
+    const __hip_bfloat16* __restrict__ src, __hip_bfloat16 (*dst)[MC_STRIDE],
+    long r0, long R, int d, int d_p) {
+  const int tr = (int)((R - r0) < MC_TR ? (R - r0) : MC_TR);
+  if ((d & 7) == 0) {
+    const int qn = d >> 3;
+    for (int i = threadIdx.x; i < tr * qn; i += MC_BLOCK) {
+      const int r = i / qn, k = (i % qn) << 3;
+      const bf16x8 v = *reinterpret_cast<const bf16x8*>(
+          &src[(r0 + r) * d + k]);
+      *reinterpret_cast<bf16x8*>(&dst[r][k]) = v;
+    }
+  } else {
+    for (int i = threadIdx.x; i < tr * d; i += MC_BLOCK) {
+      const int r = i / d, k = i % d;
+      dst[r][k] = src[(r0 + r) * d + k];
+    }
+  }
+  const __hip_bfloat16 z = __float2bfloat16(0.f);
+  const int kpad = d_p - d;
+  if (kpad > 0)
+    for (int i = threadIdx.x; i < MC_TR * kpad; i += MC_BLOCK)
+      dst[i / kpad][d + i % kpad] = z;
+  for (int i = threadIdx.x; i < (MC_TR - tr) * d; i += MC_BLOCK)
+    dst[tr + i / d][i % d] = z;
+}
+
 // stage a [dout, din] fp32 weight into LDS bf16 (optionally transposed),
-// zero-padding up to (dout_p, din_p)
+// zero-padding up to (dout_p, din_p).  The dense region loads float4
+// (4x fewer global round-trips — scalar staging measured 105us/call with
+// only ~178 WGs to hide the latency); padding cells are filled separately.
 __device__ __forceinline__ void stage_w(
     const float* __restrict__ w, __hip_bfloat16 (*ws)[MC_STRIDE], int dout,
     int din, int dout_p, int din_p, bool transpose) {
-  for (int i = threadIdx.x; i < dout_p * din_p; i += MC_BLOCK) {
-    const int o = i / din_p, k = i % din_p;
-    const float v = (o < dout && k < din) ? w[o * din + k] : 0.f;
+  if ((din & 3) == 0) {
+    const int qn = din >> 2;
+    for (int i = threadIdx.x; i < dout * qn; i += MC_BLOCK) {
+      const int o = i / qn, k = (i % qn) << 2;
+      const float4 v = reinterpret_cast<const float4*>(w)[i];
+      if (transpose) {
+        ws[k][o] = __float2bfloat16(v.x);
+        ws[k + 1][o] = __float2bfloat16(v.y);
+        ws[k + 2][o] = __float2bfloat16(v.z);
+        ws[k + 3][o] = __float2bfloat16(v.w);
+      } else {
+        __hip_bfloat162 p0, p1;
+        p0 = __hip_bfloat162{__float2bfloat16(v.x), __float2bfloat16(v.y)};
+        p1 = __hip_bfloat162{__float2bfloat16(v.z), __float2bfloat16(v.w)};
+        *reinterpret_cast<__hip_bfloat162*>(&ws[o][k]) = p0;
+        *reinterpret_cast<__hip_bfloat162*>(&ws[o][k + 2]) = p1;
+      }
+    }
+  } else {
+    for (int i = threadIdx.x; i < dout * din; i += MC_BLOCK) {
+      const int o = i / din, k = i % din;
+      const __hip_bfloat16 hv = __float2bfloat16(w[i]);
+      if (transpose)
+        ws[k][o] = hv;
+      else
+        ws[o][k] = hv;
+    }
+  }
+  // zero the padded cells (column pad for every padded row, then row pad)
+  const __hip_bfloat16 z = __float2bfloat16(0.f);
+  const int kpad = din_p - din;
+  if (kpad > 0)
+    for (int i = threadIdx.x; i < dout_p * kpad; i += MC_BLOCK) {
+      const int o = i / kpad, k = din + i % kpad;
+      if (transpose)
+        ws[k][o] = z;
+      else
+        ws[o][k] = z;
+    }
+  for (int i = threadIdx.x; i < (dout_p - dout) * din; i += MC_BLOCK) {
+    const int o = dout + i / din, k = i % din;
     if (transpose)
-      ws[k][o] = __float2bfloat16(v);
+      ws[k][o] = z;
     else
-      ws[o][k] = __float2bfloat16(v);
+      ws[o][k] = z;
   }
 }
 
@@ -110,12 +180,7 @@ npf_mlp_fwd(MlpParams p) {
 
   const int d0 = p.d[0];
   const int d0p = (d0 + 31) & ~31;
-  for (int i = threadIdx.x; i < MC_TR * d0p; i += MC_BLOCK) {
-    const int r = i / d0p, c = i % d0p;
-    a_lds[0][r][c] = (r0 + r < p.R && c < d0)
-                         ? p.x[(r0 + r) * d0 + c]
-                         : __float2bfloat16(0.f);
-  }
+  stage_x(p.x, a_lds[0], r0, p.R, d0, d0p);
 
   int cur = 0;
   for (int l = 0; l < p.L; ++l) {
@@ -187,12 +252,7 @@ npf_mlp_bwd(MlpParams p) {
   // stage dz_{L-1} = dY; store it and its db
   const int dl = p.d[L];
   const int dlp = (dl + 31) & ~31;
-  for (int i = threadIdx.x; i < MC_TR * dlp; i += MC_BLOCK) {
-    const int r = i / dlp, c = i % dlp;
-    a_lds[0][r][c] = (r0 + r < p.R && c < dl)
-                         ? p.dy[(r0 + r) * dl + c]
-                         : __float2bfloat16(0.f);
-  }
+  stage_x(p.dy, a_lds[0], r0, p.R, dl, dlp);
   __syncthreads();
   for (int i = threadIdx.x; i < MC_TR * dl; i += MC_BLOCK) {
     const int r = i / dl, c = i % dl;
