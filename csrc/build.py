@@ -25,6 +25,7 @@ def build():
     sources = [
         os.path.join(REPO, "csrc", "npf_hip", "ext.cpp"),
         os.path.join(REPO, "csrc", "npf_hip", "attn.hip"),
+        os.path.join(REPO, "csrc", "npf_hip", "attn_mfma.hip"),
         os.path.join(REPO, "csrc", "npf_hip", "setconv.hip"),
         os.path.join(REPO, "csrc", "npf_hip", "gauss_ll.hip"),
         os.path.join(REPO, "csrc", "npf_hip", "convblock.hip"),

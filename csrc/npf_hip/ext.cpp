@@ -28,6 +28,20 @@ void npf_attn_bwd_launch_bf16(const void*, const void*, const void*,
                               const void*, const float*, const void*, void*,
                               void*, void*, int, int, int, int, int, float,
                               hipStream_t);
+void npf_attn_mfma_fwd_launch_f32(const void*, const void*, const void*,
+                                  void*, float*, int, int, int, int, int,
+                                  float, hipStream_t);
+void npf_attn_mfma_fwd_launch_bf16(const void*, const void*, const void*,
+                                   void*, float*, int, int, int, int, int,
+                                   float, hipStream_t);
+void npf_attn_mfma_bwd_launch_f32(const void*, const void*, const void*,
+                                  const void*, const float*, const void*,
+                                  float*, void*, void*, void*, int, int, int,
+                                  int, int, float, hipStream_t);
+void npf_attn_mfma_bwd_launch_bf16(const void*, const void*, const void*,
+                                   const void*, const float*, const void*,
+                                   float*, void*, void*, void*, int, int,
+                                   int, int, int, float, hipStream_t);
 void npf_setconv_fwd_launch(const float*, const float*, const float*, float*,
                             int, int, int, int, const float*, hipStream_t);
 void npf_setconv_bwd_launch(const float*, const float*, const float*,
@@ -106,15 +120,18 @@ std::tuple<torch::Tensor, torch::Tensor> attn_fwd(torch::Tensor q,
   auto out = torch::empty({N, Q, Dv}, q.options());
   auto lse = torch::empty({N, Q}, q.options().dtype(torch::kFloat32));
   auto stream = at::hip::getCurrentHIPStream();
+  // large-K regime -> MFMA flash kernel (attn_mfma.hip); tiny K (1D models)
+  // -> thread-per-query VALU kernel where MFMA tiles would be mostly padding
+  const bool mfma = K >= 96;
   if (is_bf16(q)) {
-    npf_attn_fwd_launch_bf16(q.data_ptr(), k.data_ptr(), v.data_ptr(),
-                             out.data_ptr(), lse.data_ptr<float>(), N, Q, K, D,
-                             Dv, (float)scale, stream);
+    (mfma ? npf_attn_mfma_fwd_launch_bf16 : npf_attn_fwd_launch_bf16)(
+        q.data_ptr(), k.data_ptr(), v.data_ptr(), out.data_ptr(),
+        lse.data_ptr<float>(), N, Q, K, D, Dv, (float)scale, stream);
   } else {
     TORCH_CHECK(q.scalar_type() == torch::kFloat32, "fp32 or bf16 only");
-    npf_attn_fwd_launch_f32(q.data_ptr(), k.data_ptr(), v.data_ptr(),
-                            out.data_ptr(), lse.data_ptr<float>(), N, Q, K, D,
-                            Dv, (float)scale, stream);
+    (mfma ? npf_attn_mfma_fwd_launch_f32 : npf_attn_fwd_launch_f32)(
+        q.data_ptr(), k.data_ptr(), v.data_ptr(), out.data_ptr(),
+        lse.data_ptr<float>(), N, Q, K, D, Dv, (float)scale, stream);
   }
   return {out, lse};
 }
@@ -134,7 +151,22 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> attn_bwd(
   auto dk = torch::empty_like(k);
   auto dv = torch::empty_like(v);
   auto stream = at::hip::getCurrentHIPStream();
-  if (is_bf16(q)) {
+  const bool mfma = K >= 96;
+  if (mfma) {
+    auto delta = torch::empty({N, Q}, lse.options());
+    if (is_bf16(q))
+      npf_attn_mfma_bwd_launch_bf16(
+          q.data_ptr(), k.data_ptr(), v.data_ptr(), out.data_ptr(),
+          lse.data_ptr<float>(), dout.data_ptr(), delta.data_ptr<float>(),
+          dq.data_ptr(), dk.data_ptr(), dv.data_ptr(), N, Q, K, D, Dv,
+          (float)scale, stream);
+    else
+      npf_attn_mfma_bwd_launch_f32(
+          q.data_ptr(), k.data_ptr(), v.data_ptr(), out.data_ptr(),
+          lse.data_ptr<float>(), dout.data_ptr(), delta.data_ptr<float>(),
+          dq.data_ptr(), dk.data_ptr(), dv.data_ptr(), N, Q, K, D, Dv,
+          (float)scale, stream);
+  } else if (is_bf16(q)) {
     npf_attn_bwd_launch_bf16(q.data_ptr(), k.data_ptr(), v.data_ptr(),
                              out.data_ptr(), lse.data_ptr<float>(),
                              dout.data_ptr(), dq.data_ptr(), dk.data_ptr(),

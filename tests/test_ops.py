@@ -113,8 +113,12 @@ class TestHIPKernels:
         if not has_extension():
             pytest.fail("HIP extension not built/loadable on a GPU box")
 
-    @pytest.mark.parametrize("B,K,Q,D", [(8, 13, 128, 16), (32, 50, 128, 16),
-                                         (256, 300, 1024, 16)])
+    @pytest.mark.parametrize("B,K,Q,D", [
+        (8, 13, 128, 16), (32, 50, 128, 16),      # VALU path (K < 96)
+        (256, 300, 1024, 16),                      # MFMA path
+        (8, 96, 64, 16), (4, 130, 50, 16),        # MFMA boundary / small Q
+        (16, 307, 307, 16), (8, 1024, 1024, 16),  # self-attn shapes
+    ])
     def test_attention_fwd_bwd(self, B, K, Q, D):
         g = torch.Generator(device="cuda").manual_seed(0)
         k = torch.randn(B, K, D, device="cuda", generator=g, requires_grad=True)
