@@ -120,9 +120,10 @@ std::tuple<torch::Tensor, torch::Tensor> attn_fwd(torch::Tensor q,
   auto out = torch::empty({N, Q, Dv}, q.options());
   auto lse = torch::empty({N, Q}, q.options().dtype(torch::kFloat32));
   auto stream = at::hip::getCurrentHIPStream();
-  // large-K regime -> MFMA flash kernel (attn_mfma.hip); tiny K (1D models)
-  // -> thread-per-query VALU kernel where MFMA tiles would be mostly padding
-  const bool mfma = K >= 96;
+  // large-K bf16 -> MFMA flash kernel (attn_mfma.hip).  fp32 inputs keep
+  // the VALU kernel (exact fp32 math — MFMA would silently round operands
+  // to bf16); tiny K also stays VALU (MFMA tiles would be mostly padding).
+  const bool mfma = K >= 96 && is_bf16(q);
   if (is_bf16(q)) {
     (mfma ? npf_attn_mfma_fwd_launch_bf16 : npf_attn_fwd_launch_bf16)(
         q.data_ptr(), k.data_ptr(), v.data_ptr(), out.data_ptr(),
@@ -151,7 +152,7 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> attn_bwd(
   auto dk = torch::empty_like(k);
   auto dv = torch::empty_like(v);
   auto stream = at::hip::getCurrentHIPStream();
-  const bool mfma = K >= 96;
+  const bool mfma = K >= 96 && is_bf16(q);
   if (mfma) {
     auto delta = torch::empty({N, Q}, lse.options());
     if (is_bf16(q))

@@ -113,31 +113,43 @@ class TestHIPKernels:
         if not has_extension():
             pytest.fail("HIP extension not built/loadable on a GPU box")
 
-    @pytest.mark.parametrize("B,K,Q,D", [
-        (8, 13, 128, 16), (32, 50, 128, 16),      # VALU path (K < 96)
-        (256, 300, 1024, 16),                      # MFMA path
-        (8, 96, 64, 16), (4, 130, 50, 16),        # MFMA boundary / small Q
-        (16, 307, 307, 16), (8, 1024, 1024, 16),  # self-attn shapes
+    @pytest.mark.parametrize("B,K,Q,D,dtype", [
+        # fp32 -> exact VALU kernel, tight tolerance
+        (8, 13, 128, 16, "f32"), (32, 50, 128, 16, "f32"),
+        (256, 300, 1024, 16, "f32"), (8, 96, 64, 16, "f32"),
+        # bf16 large-K -> MFMA flash kernel, bf16-level tolerance
+        (8, 96, 64, 16, "bf16"), (4, 130, 50, 16, "bf16"),
+        (16, 307, 307, 16, "bf16"), (256, 300, 1024, 16, "bf16"),
+        (8, 1024, 1024, 16, "bf16"),
+        # bf16 small-K -> VALU kernel on bf16 inputs
+        (32, 50, 128, 16, "bf16"),
     ])
-    def test_attention_fwd_bwd(self, B, K, Q, D):
+    def test_attention_fwd_bwd(self, B, K, Q, D, dtype):
+        td = torch.float32 if dtype == "f32" else torch.bfloat16
+        atol_o, atol_g = (2e-4, 2e-3) if dtype == "f32" else (3e-2, 6e-2)
         g = torch.Generator(device="cuda").manual_seed(0)
-        k = torch.randn(B, K, D, device="cuda", generator=g, requires_grad=True)
-        q = torch.randn(B, Q, D, device="cuda", generator=g, requires_grad=True)
-        v = torch.randn(B, K, D, device="cuda", generator=g, requires_grad=True)
+        k = torch.randn(B, K, D, device="cuda", generator=g, dtype=td,
+                        requires_grad=True)
+        q = torch.randn(B, Q, D, device="cuda", generator=g, dtype=td,
+                        requires_grad=True)
+        v = torch.randn(B, K, D, device="cuda", generator=g, dtype=td,
+                        requires_grad=True)
         scale = 1 / math.sqrt(D)
         out = F_ops.attention_qkv(k, q, v, scale)
-        k0 = k.detach().cpu().requires_grad_(True)
-        q0 = q.detach().cpu().requires_grad_(True)
-        v0 = v.detach().cpu().requires_grad_(True)
+        k0 = k.detach().cpu().float().requires_grad_(True)
+        q0 = q.detach().cpu().float().requires_grad_(True)
+        v0 = v.detach().cpu().float().requires_grad_(True)
         ref = _attn_oracle(k0, q0, v0, scale)
-        assert torch.allclose(out.cpu(), ref, atol=2e-4), (
-            (out.cpu() - ref).abs().max()
+        assert torch.allclose(out.cpu().float(), ref, atol=atol_o), (
+            (out.cpu().float() - ref).abs().max()
         )
         dout = torch.randn_like(out)
         out.backward(dout)
-        ref.backward(dout.cpu())
+        ref.backward(dout.cpu().float())
         for a, b in [(k.grad, k0.grad), (q.grad, q0.grad), (v.grad, v0.grad)]:
-            assert torch.allclose(a.cpu(), b, atol=2e-3), (a.cpu() - b).abs().max()
+            assert torch.allclose(a.cpu().float(), b, atol=atol_g), (
+                (a.cpu().float() - b).abs().max()
+            )
 
     @pytest.mark.parametrize("B,K,Q,C", [(4, 17, 33, 1), (32, 50, 192, 1),
                                          (64, 192, 128, 128)])
