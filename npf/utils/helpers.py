@@ -324,3 +324,26 @@ class CircularPad2d(nn.Module):
 
     def forward(self, x):
         return F.pad(x, (self.padding,) * 4, mode="circular")
+
+
+class BackwardPDB(torch.autograd.Function):
+    """Autograd identity that drops into pdb when its gradient is non-finite
+    (reference helpers.py:417-436 — debugging hook for exploding losses)."""
+
+    @staticmethod
+    def forward(ctx, x, name="debugger"):
+        ctx.name = name
+        return x
+
+    @staticmethod
+    def backward(ctx, grad_output):
+        if not torch.isfinite(grad_output).all():
+            import pdb
+
+            pdb.set_trace()  # noqa: T100  (intentional: debug hook)
+        return grad_output, None
+
+
+def backward_pdb(x, name="debugger"):
+    """Insert a BackwardPDB probe on `x`."""
+    return BackwardPDB.apply(x, name)
