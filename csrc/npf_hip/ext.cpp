@@ -52,8 +52,9 @@ void npf_gauss_ll_fwd_launch(const float*, const float*, const float*, float*,
 void npf_gauss_ll_bwd_launch(const float*, const float*, const float*,
                              const float*, float*, float*, long long,
                              long long, hipStream_t);
-void npf_cb_stats_launch(const float*, float*, float*, float*, float*, float*,
-                         int, int, int, float, float, hipStream_t);
+void npf_cb_stats_launch(const float*, float*, float*, float*, float*,
+                         float*, float*, int, int, int, float, float,
+                         hipStream_t);
 void npf_cb_fwd_launch(const float*, const float*, const float*, const float*,
                        const float*, const float*, const float*, const float*,
                        float*, int, int, int, int, hipStream_t);
@@ -271,12 +272,14 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> convblock_stats(
   auto mean = torch::empty({C}, x.options());
   auto rstd = torch::empty({C}, x.options());
   auto save_var = torch::empty({C}, x.options());
+  auto sums = torch::zeros({2 * C}, x.options());  // (sum ; sumsq) workspace
   auto stream = at::hip::getCurrentHIPStream();
   float* rm = running_mean.defined() ? running_mean.data_ptr<float>() : nullptr;
   float* rv = running_var.defined() ? running_var.data_ptr<float>() : nullptr;
-  npf_cb_stats_launch(x.data_ptr<float>(), mean.data_ptr<float>(),
-                      rstd.data_ptr<float>(), save_var.data_ptr<float>(), rm,
-                      rv, N, C, L, (float)eps, (float)momentum, stream);
+  npf_cb_stats_launch(x.data_ptr<float>(), sums.data_ptr<float>(),
+                      mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                      save_var.data_ptr<float>(), rm, rv, N, C, L, (float)eps,
+                      (float)momentum, stream);
   return {mean, rstd, save_var};
 }
 
