@@ -18,12 +18,14 @@
 #define CB2_BLOCK 256
 #define CB2_MAX_K 13
 
-extern "C" __global__ void __launch_bounds__(CB2_BLOCK)
+template <int KT>
+__global__ void __launch_bounds__(CB2_BLOCK)
 npf_cb2d_fwd(const float* __restrict__ x, const float* __restrict__ res,
              const float* __restrict__ w, const float* __restrict__ bias,
              const float* __restrict__ gamma, const float* __restrict__ beta,
              const float* __restrict__ mean, const float* __restrict__ rstd,
-             float* __restrict__ y, int N, int C, int H, int W, int K) {
+             float* __restrict__ y, int N, int C, int H, int W, int K_rt) {
+  const int K = (KT > 0) ? KT : K_rt;
   extern __shared__ float smem[];
   const int pad = K / 2;
   const int HP = H + 2 * pad, WP = W + 2 * pad;
@@ -56,9 +58,10 @@ npf_cb2d_fwd(const float* __restrict__ x, const float* __restrict__ res,
   for (int i = threadIdx.x; i < H * W; i += CB2_BLOCK) {
     const int r = i / W, col = i % W;
     float acc = b;
+    #pragma unroll
     for (int kr = 0; kr < K; ++kr) {
       const float* arow = a + (r + kr) * WP + col;
-      #pragma unroll 3
+      #pragma unroll
       for (int kc = 0; kc < K; ++kc) acc += ws[kr * K + kc] * arow[kc];
     }
     if (rpl != nullptr) acc += rpl[i];
@@ -68,7 +71,8 @@ npf_cb2d_fwd(const float* __restrict__ x, const float* __restrict__ res,
 
 // backward stencil: dact (stored), dW/db block-reduced then global atomics,
 // BN channel partial sums (global atomics)
-extern "C" __global__ void __launch_bounds__(CB2_BLOCK)
+template <int KT>
+__global__ void __launch_bounds__(CB2_BLOCK)
 npf_cb2d_bwd_dact(const float* __restrict__ x, const float* __restrict__ w,
                   const float* __restrict__ dy, const float* __restrict__ gamma,
                   const float* __restrict__ beta, const float* __restrict__ mean,
@@ -77,7 +81,8 @@ npf_cb2d_bwd_dact(const float* __restrict__ x, const float* __restrict__ w,
                   float* __restrict__ sum_dxhat,
                   float* __restrict__ sum_dxhat_xhat,
                   float* __restrict__ dgamma, float* __restrict__ dbeta,
-                  int N, int C, int H, int W, int K) {
+                  int N, int C, int H, int W, int K_rt) {
+  const int K = (KT > 0) ? KT : K_rt;
   extern __shared__ float smem[];
   const int pad = K / 2;
   const int HP = H + 2 * pad, WP = W + 2 * pad;
@@ -117,9 +122,10 @@ npf_cb2d_bwd_dact(const float* __restrict__ x, const float* __restrict__ w,
     dbp += dyl;
     // transposed conv: da[p] = sum_k w[k] dY[p + pad - k]
     float da = 0.f;
+    #pragma unroll
     for (int kr = 0; kr < K; ++kr) {
       const float* drow = dys + (r + K - 1 - kr) * WP + col;
-      #pragma unroll 3
+      #pragma unroll
       for (int kc = 0; kc < K; ++kc) da += ws[kr * K + kc] * drow[K - 1 - kc];
     }
     const float act = a[(r + pad) * WP + col + pad];
@@ -189,9 +195,24 @@ extern "C" void npf_cb2d_fwd_launch(const float* x, const float* res,
   const int pad = K / 2;
   const size_t smem =
       ((size_t)(H + 2 * pad) * (W + 2 * pad) + K * K) * sizeof(float);
-  hipLaunchKernelGGL(npf_cb2d_fwd, dim3((unsigned)N * C), dim3(CB2_BLOCK),
-                     smem, stream, x, res, w, bias, gamma, beta, mean, rstd, y,
-                     N, C, H, W, K);
+  const dim3 grid((unsigned)N * C), blk(CB2_BLOCK);
+  switch (K) {  // compile-time stencil: full unroll + hoisted addresses
+    case 5:
+      hipLaunchKernelGGL((npf_cb2d_fwd<5>), grid, blk, smem, stream, x, res,
+                         w, bias, gamma, beta, mean, rstd, y, N, C, H, W, K);
+      break;
+    case 9:
+      hipLaunchKernelGGL((npf_cb2d_fwd<9>), grid, blk, smem, stream, x, res,
+                         w, bias, gamma, beta, mean, rstd, y, N, C, H, W, K);
+      break;
+    case 11:
+      hipLaunchKernelGGL((npf_cb2d_fwd<11>), grid, blk, smem, stream, x, res,
+                         w, bias, gamma, beta, mean, rstd, y, N, C, H, W, K);
+      break;
+    default:
+      hipLaunchKernelGGL((npf_cb2d_fwd<0>), grid, blk, smem, stream, x, res,
+                         w, bias, gamma, beta, mean, rstd, y, N, C, H, W, K);
+  }
 }
 
 extern "C" void npf_cb2d_bwd_dact_launch(
@@ -203,8 +224,30 @@ extern "C" void npf_cb2d_bwd_dact_launch(
   const int pad = K / 2;
   const size_t smem =
       (2 * (size_t)(H + 2 * pad) * (W + 2 * pad) + K * K) * sizeof(float);
-  hipLaunchKernelGGL(npf_cb2d_bwd_dact, dim3((unsigned)N * C),
-                     dim3(CB2_BLOCK), smem, stream, x, w, dy, gamma, beta,
-                     mean, rstd, dact, dw, db, sum_dxhat, sum_dxhat_xhat,
-                     dgamma, dbeta, N, C, H, W, K);
+  const dim3 grid((unsigned)N * C), blk(CB2_BLOCK);
+  switch (K) {
+    case 5:
+      hipLaunchKernelGGL((npf_cb2d_bwd_dact<5>), grid, blk, smem, stream, x,
+                         w, dy, gamma, beta, mean, rstd, dact, dw, db,
+                         sum_dxhat, sum_dxhat_xhat, dgamma, dbeta, N, C, H, W,
+                         K);
+      break;
+    case 9:
+      hipLaunchKernelGGL((npf_cb2d_bwd_dact<9>), grid, blk, smem, stream, x,
+                         w, dy, gamma, beta, mean, rstd, dact, dw, db,
+                         sum_dxhat, sum_dxhat_xhat, dgamma, dbeta, N, C, H, W,
+                         K);
+      break;
+    case 11:
+      hipLaunchKernelGGL((npf_cb2d_bwd_dact<11>), grid, blk, smem, stream, x,
+                         w, dy, gamma, beta, mean, rstd, dact, dw, db,
+                         sum_dxhat, sum_dxhat_xhat, dgamma, dbeta, N, C, H, W,
+                         K);
+      break;
+    default:
+      hipLaunchKernelGGL((npf_cb2d_bwd_dact<0>), grid, blk, smem, stream, x,
+                         w, dy, gamma, beta, mean, rstd, dact, dw, db,
+                         sum_dxhat, sum_dxhat_xhat, dgamma, dbeta, N, C, H, W,
+                         K);
+  }
 }
