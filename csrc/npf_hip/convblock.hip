@@ -40,8 +40,9 @@
 // then a tiny finalize kernel.
 #define CB_STATS_SLICE 16   // batch rows per partial-reduction workgroup
 
-extern "C" __global__ void __launch_bounds__(CB_BLOCK)
-npf_cb_stats_partial(const float* __restrict__ x, float* __restrict__ sums,
+template <typename T>
+__global__ void __launch_bounds__(CB_BLOCK)
+npf_cb_stats_partial(const T* __restrict__ x, float* __restrict__ sums,
                      int N, int C, int L) {
   __shared__ float red[16];
   const int c = blockIdx.x % C;
@@ -50,7 +51,7 @@ npf_cb_stats_partial(const float* __restrict__ x, float* __restrict__ sums,
   float s = 0.f, s2 = 0.f;
   for (long i = threadIdx.x; i < (long)nt * L; i += CB_BLOCK) {
     const long n = n0 + i / L, l = i % L;
-    const float v = x[(n * C + c) * (long)L + l];
+    const float v = ldf(x + (n * C + c) * (long)L + l);
     s += v;
     s2 += v * v;
   }
@@ -237,12 +238,13 @@ npf_cb_bwd_dact(const float* __restrict__ x, const float* __restrict__ w,
 }
 
 // B2: dx = bn_backward(dact) (+ dY residual)
-extern "C" __global__ void __launch_bounds__(CB_BLOCK)
-npf_cb_bwd_dx(const float* __restrict__ x, const float* __restrict__ dact,
-              const float* __restrict__ dy, const float* __restrict__ gamma,
+template <typename T>
+__global__ void __launch_bounds__(CB_BLOCK)
+npf_cb_bwd_dx(const T* __restrict__ x, const T* __restrict__ dact,
+              const T* __restrict__ dy, const float* __restrict__ gamma,
               const float* __restrict__ mean, const float* __restrict__ rstd,
               const float* __restrict__ sum_dxhat,
-              const float* __restrict__ sum_dxhat_xhat, float* __restrict__ dx,
+              const float* __restrict__ sum_dxhat_xhat, T* __restrict__ dx,
               int N, int C, int L, int training) {
   const int n = blockIdx.x / C;
   const int c = blockIdx.x % C;
@@ -257,17 +259,17 @@ npf_cb_bwd_dx(const float* __restrict__ x, const float* __restrict__ dact,
   for (int l = threadIdx.x; l < L; l += CB_BLOCK) {
     float d;
     if (has_bn) {
-      const float dxh = dact[off + l] * g;
+      const float dxh = ldf(dact + off + l) * g;
       if (training) {
-        const float xhat = (x[off + l] - mu) * rs;
+        const float xhat = (ldf(x + off + l) - mu) * rs;
         d = rs * (dxh - mean_dxh - xhat * mean_dxh_xh);
       } else {
         d = rs * dxh;  // eval mode: stats are constants
       }
     } else {
-      d = dact[off + l];
+      d = ldf(dact + off + l);
     }
-    dx[off + l] = d;
+    stf(dx + off + l, d);
   }
 }
 
@@ -275,14 +277,21 @@ npf_cb_bwd_dx(const float* __restrict__ x, const float* __restrict__ dact,
 // host launchers
 // ---------------------------------------------------------------------------
 
-extern "C" void npf_cb_stats_launch(const float* x, float* sums_ws,
-                                    float* mean, float* rstd, float* save_var,
-                                    float* running_mean, float* running_var,
-                                    int N, int C, int L, float eps,
-                                    float momentum, hipStream_t stream) {
+extern "C" void npf_cb_stats_launch(const void* x, int is_bf16,
+                                    float* sums_ws, float* mean, float* rstd,
+                                    float* save_var, float* running_mean,
+                                    float* running_var, int N, int C, int L,
+                                    float eps, float momentum,
+                                    hipStream_t stream) {
   const unsigned slices = (unsigned)((N + CB_STATS_SLICE - 1) / CB_STATS_SLICE);
-  hipLaunchKernelGGL(npf_cb_stats_partial, dim3(slices * C), dim3(CB_BLOCK),
-                     0, stream, x, sums_ws, N, C, L);
+  if (is_bf16)
+    hipLaunchKernelGGL((npf_cb_stats_partial<__hip_bfloat16>),
+                       dim3(slices * C), dim3(CB_BLOCK), 0, stream,
+                       (const __hip_bfloat16*)x, sums_ws, N, C, L);
+  else
+    hipLaunchKernelGGL((npf_cb_stats_partial<float>), dim3(slices * C),
+                       dim3(CB_BLOCK), 0, stream, (const float*)x, sums_ws,
+                       N, C, L);
   hipLaunchKernelGGL(npf_cb_stats_finalize, dim3((C + 255) / 256), dim3(256),
                      0, stream, sums_ws, mean, rstd, save_var, running_mean,
                      running_var, N, C, L, eps, momentum);
@@ -315,21 +324,31 @@ extern "C" void npf_cb_bwd_launch(const float* x, const float* w,
                      smem, stream, x, w, dy, gamma, beta, mean, rstd, dact,
                      dw, db, sum_dxhat, sum_dxhat_xhat, dgamma, dbeta, N, C,
                      L, K);
-  hipLaunchKernelGGL(npf_cb_bwd_dx, dim3((unsigned)N * C), dim3(CB_BLOCK), 0,
-                     stream, x, dact, dy, gamma, mean, rstd, sum_dxhat,
-                     sum_dxhat_xhat, dx, N, C, L, training);
+  hipLaunchKernelGGL((npf_cb_bwd_dx<float>), dim3((unsigned)N * C),
+                     dim3(CB_BLOCK), 0, stream, x, dact, dy, gamma, mean,
+                     rstd, sum_dxhat, sum_dxhat_xhat, dx, N, C, L, training);
 }
 
 // standalone launcher for the elementwise BN-backward kernel (shared by the
-// 2D block, which flattens L = H*W)
-extern "C" void npf_cb_bwd_dx_launch(const float* x, const float* dact,
-                                     const float* dy, const float* gamma,
+// 2D block, which flattens L = H*W); dtype-dispatched
+extern "C" void npf_cb_bwd_dx_launch(const void* x, const void* dact,
+                                     const void* dy, const float* gamma,
                                      const float* mean, const float* rstd,
                                      const float* sum_dxhat,
-                                     const float* sum_dxhat_xhat, float* dx,
+                                     const float* sum_dxhat_xhat, void* dx,
                                      int N, int C, int L, int training,
-                                     hipStream_t stream) {
-  hipLaunchKernelGGL(npf_cb_bwd_dx, dim3((unsigned)N * C), dim3(CB_BLOCK), 0,
-                     stream, x, dact, dy, gamma, mean, rstd, sum_dxhat,
-                     sum_dxhat_xhat, dx, N, C, L, training);
+                                     int is_bf16, hipStream_t stream) {
+  if (is_bf16)
+    hipLaunchKernelGGL((npf_cb_bwd_dx<__hip_bfloat16>),
+                       dim3((unsigned)N * C), dim3(CB_BLOCK), 0, stream,
+                       (const __hip_bfloat16*)x, (const __hip_bfloat16*)dact,
+                       (const __hip_bfloat16*)dy, gamma, mean, rstd,
+                       sum_dxhat, sum_dxhat_xhat, (__hip_bfloat16*)dx, N, C,
+                       L, training);
+  else
+    hipLaunchKernelGGL((npf_cb_bwd_dx<float>), dim3((unsigned)N * C),
+                       dim3(CB_BLOCK), 0, stream, (const float*)x,
+                       (const float*)dact, (const float*)dy, gamma, mean,
+                       rstd, sum_dxhat, sum_dxhat_xhat, (float*)dx, N, C, L,
+                       training);
 }

@@ -18,13 +18,13 @@
 #define CB2_BLOCK 256
 #define CB2_MAX_K 13
 
-template <int KT>
+template <typename T, int KT>
 __global__ void __launch_bounds__(CB2_BLOCK)
-npf_cb2d_fwd(const float* __restrict__ x, const float* __restrict__ res,
+npf_cb2d_fwd(const T* __restrict__ x, const T* __restrict__ res,
              const float* __restrict__ w, const float* __restrict__ bias,
              const float* __restrict__ gamma, const float* __restrict__ beta,
              const float* __restrict__ mean, const float* __restrict__ rstd,
-             float* __restrict__ y, int N, int C, int H, int W, int K_rt) {
+             T* __restrict__ y, int N, int C, int H, int W, int K_rt) {
   const int K = (KT > 0) ? KT : K_rt;
   extern __shared__ float smem[];
   const int pad = K / 2;
@@ -34,7 +34,7 @@ npf_cb2d_fwd(const float* __restrict__ x, const float* __restrict__ res,
 
   const int n = blockIdx.x / C;
   const int c = blockIdx.x % C;
-  const float* xpl = x + ((long)n * C + c) * H * W;
+  const T* xpl = x + ((long)n * C + c) * H * W;
 
   const bool has_bn = gamma != nullptr;
   const float mu = has_bn ? mean[c] : 0.f;
@@ -46,14 +46,14 @@ npf_cb2d_fwd(const float* __restrict__ x, const float* __restrict__ res,
     const int r = i / WP - pad, col = i % WP - pad;
     float v = 0.f;
     if (r >= 0 && r < H && col >= 0 && col < W)
-      v = fmaxf((xpl[r * W + col] - mu) * gscale + gshift, 0.f);
+      v = fmaxf((ldf(xpl + r * W + col) - mu) * gscale + gshift, 0.f);
     a[i] = v;
   }
   __syncthreads();
 
   const float b = (bias != nullptr) ? bias[c] : 0.f;
-  float* ypl = y + ((long)n * C + c) * H * W;
-  const float* rpl =
+  T* ypl = y + ((long)n * C + c) * H * W;
+  const T* rpl =
       (res != nullptr) ? res + ((long)n * C + c) * H * W : nullptr;
   for (int i = threadIdx.x; i < H * W; i += CB2_BLOCK) {
     const int r = i / W, col = i % W;
@@ -64,19 +64,19 @@ npf_cb2d_fwd(const float* __restrict__ x, const float* __restrict__ res,
       #pragma unroll
       for (int kc = 0; kc < K; ++kc) acc += ws[kr * K + kc] * arow[kc];
     }
-    if (rpl != nullptr) acc += rpl[i];
-    ypl[i] = acc;
+    if (rpl != nullptr) acc += ldf(rpl + i);
+    stf(ypl + i, acc);
   }
 }
 
 // backward stencil: dact (stored), dW/db block-reduced then global atomics,
 // BN channel partial sums (global atomics)
-template <int KT>
+template <typename T, int KT>
 __global__ void __launch_bounds__(CB2_BLOCK)
-npf_cb2d_bwd_dact(const float* __restrict__ x, const float* __restrict__ w,
-                  const float* __restrict__ dy, const float* __restrict__ gamma,
+npf_cb2d_bwd_dact(const T* __restrict__ x, const float* __restrict__ w,
+                  const T* __restrict__ dy, const float* __restrict__ gamma,
                   const float* __restrict__ beta, const float* __restrict__ mean,
-                  const float* __restrict__ rstd, float* __restrict__ dact,
+                  const float* __restrict__ rstd, T* __restrict__ dact,
                   float* __restrict__ dw, float* __restrict__ db,
                   float* __restrict__ sum_dxhat,
                   float* __restrict__ sum_dxhat_xhat,
@@ -93,8 +93,8 @@ npf_cb2d_bwd_dact(const float* __restrict__ x, const float* __restrict__ w,
 
   const int n = blockIdx.x / C;
   const int c = blockIdx.x % C;
-  const float* xpl = x + ((long)n * C + c) * H * W;
-  const float* dypl = dy + ((long)n * C + c) * H * W;
+  const T* xpl = x + ((long)n * C + c) * H * W;
+  const T* dypl = dy + ((long)n * C + c) * H * W;
 
   const bool has_bn = gamma != nullptr;
   const float mu = has_bn ? mean[c] : 0.f;
@@ -107,18 +107,18 @@ npf_cb2d_bwd_dact(const float* __restrict__ x, const float* __restrict__ w,
   for (int i = threadIdx.x; i < HP * WP; i += CB2_BLOCK) {
     const int r = i / WP - pad, col = i % WP - pad;
     const bool in = (r >= 0 && r < H && col >= 0 && col < W);
-    dys[i] = in ? dypl[r * W + col] : 0.f;
+    dys[i] = in ? ldf(dypl + r * W + col) : 0.f;
     float v = 0.f;
-    if (in) v = fmaxf((xpl[r * W + col] - mu) * gscale + gshift, 0.f);
+    if (in) v = fmaxf((ldf(xpl + r * W + col) - mu) * gscale + gshift, 0.f);
     a[i] = v;
   }
   __syncthreads();
 
-  float* dactpl = dact + ((long)n * C + c) * H * W;
+  T* dactpl = dact + ((long)n * C + c) * H * W;
   float s_dxhat = 0.f, s_dxhat_xhat = 0.f, s_dg = 0.f, s_db = 0.f, dbp = 0.f;
   for (int i = threadIdx.x; i < H * W; i += CB2_BLOCK) {
     const int r = i / W, col = i % W;
-    const float dyl = dypl[i];
+    const float dyl = ldf(dypl + i);
     dbp += dyl;
     // transposed conv: da[p] = sum_k w[k] dY[p + pad - k]
     float da = 0.f;
@@ -130,9 +130,9 @@ npf_cb2d_bwd_dact(const float* __restrict__ x, const float* __restrict__ w,
     }
     const float act = a[(r + pad) * WP + col + pad];
     const float dr = (act > 0.f) ? da : 0.f;
-    dactpl[i] = dr;
+    stf(dactpl + i, dr);
     if (has_bn) {
-      const float xhat = (xpl[i] - mu) * rs;
+      const float xhat = (ldf(xpl + i) - mu) * rs;
       const float dxh = dr * gm;
       s_dxhat += dxh;
       s_dxhat_xhat += dxh * xhat;
@@ -186,68 +186,82 @@ npf_cb2d_bwd_dact(const float* __restrict__ x, const float* __restrict__ w,
 // L = H*W, launched from ext.cpp)
 // ---------------------------------------------------------------------------
 
-extern "C" void npf_cb2d_fwd_launch(const float* x, const float* res,
-                                    const float* w, const float* bias,
-                                    const float* gamma, const float* beta,
-                                    const float* mean, const float* rstd,
-                                    float* y, int N, int C, int H, int W,
-                                    int K, hipStream_t stream) {
+template <typename T>
+static void cb2d_fwd_dispatch(const void* x, const void* res, const float* w,
+                              const float* bias, const float* gamma,
+                              const float* beta, const float* mean,
+                              const float* rstd, void* y, int N, int C, int H,
+                              int W, int K, hipStream_t stream) {
   const int pad = K / 2;
   const size_t smem =
       ((size_t)(H + 2 * pad) * (W + 2 * pad) + K * K) * sizeof(float);
   const dim3 grid((unsigned)N * C), blk(CB2_BLOCK);
+#define CB2D_FWD(KT)                                                        \
+  hipLaunchKernelGGL((npf_cb2d_fwd<T, KT>), grid, blk, smem, stream,        \
+                     (const T*)x, (const T*)res, w, bias, gamma, beta,      \
+                     mean, rstd, (T*)y, N, C, H, W, K)
   switch (K) {  // compile-time stencil: full unroll + hoisted addresses
-    case 5:
-      hipLaunchKernelGGL((npf_cb2d_fwd<5>), grid, blk, smem, stream, x, res,
-                         w, bias, gamma, beta, mean, rstd, y, N, C, H, W, K);
-      break;
-    case 9:
-      hipLaunchKernelGGL((npf_cb2d_fwd<9>), grid, blk, smem, stream, x, res,
-                         w, bias, gamma, beta, mean, rstd, y, N, C, H, W, K);
-      break;
-    case 11:
-      hipLaunchKernelGGL((npf_cb2d_fwd<11>), grid, blk, smem, stream, x, res,
-                         w, bias, gamma, beta, mean, rstd, y, N, C, H, W, K);
-      break;
-    default:
-      hipLaunchKernelGGL((npf_cb2d_fwd<0>), grid, blk, smem, stream, x, res,
-                         w, bias, gamma, beta, mean, rstd, y, N, C, H, W, K);
+    case 5: CB2D_FWD(5); break;
+    case 9: CB2D_FWD(9); break;
+    case 11: CB2D_FWD(11); break;
+    default: CB2D_FWD(0);
   }
+#undef CB2D_FWD
 }
 
-extern "C" void npf_cb2d_bwd_dact_launch(
-    const float* x, const float* w, const float* dy, const float* gamma,
-    const float* beta, const float* mean, const float* rstd, float* dact,
-    float* dw, float* db, float* sum_dxhat, float* sum_dxhat_xhat,
-    float* dgamma, float* dbeta, int N, int C, int H, int W, int K,
-    hipStream_t stream) {
+extern "C" void npf_cb2d_fwd_launch(const void* x, const void* res,
+                                    const float* w, const float* bias,
+                                    const float* gamma, const float* beta,
+                                    const float* mean, const float* rstd,
+                                    void* y, int N, int C, int H, int W,
+                                    int K, int is_bf16, hipStream_t stream) {
+  if (is_bf16)
+    cb2d_fwd_dispatch<__hip_bfloat16>(x, res, w, bias, gamma, beta, mean,
+                                      rstd, y, N, C, H, W, K, stream);
+  else
+    cb2d_fwd_dispatch<float>(x, res, w, bias, gamma, beta, mean, rstd, y, N,
+                             C, H, W, K, stream);
+}
+
+template <typename T>
+static void cb2d_bwd_dispatch(const void* x, const float* w, const void* dy,
+                              const float* gamma, const float* beta,
+                              const float* mean, const float* rstd,
+                              void* dact, float* dw, float* db,
+                              float* sum_dxhat, float* sum_dxhat_xhat,
+                              float* dgamma, float* dbeta, int N, int C,
+                              int H, int W, int K, hipStream_t stream) {
   const int pad = K / 2;
   const size_t smem =
       (2 * (size_t)(H + 2 * pad) * (W + 2 * pad) + K * K) * sizeof(float);
   const dim3 grid((unsigned)N * C), blk(CB2_BLOCK);
+#define CB2D_BWD(KT)                                                        \
+  hipLaunchKernelGGL((npf_cb2d_bwd_dact<T, KT>), grid, blk, smem, stream,   \
+                     (const T*)x, w, (const T*)dy, gamma, beta, mean, rstd, \
+                     (T*)dact, dw, db, sum_dxhat, sum_dxhat_xhat, dgamma,   \
+                     dbeta, N, C, H, W, K)
   switch (K) {
-    case 5:
-      hipLaunchKernelGGL((npf_cb2d_bwd_dact<5>), grid, blk, smem, stream, x,
-                         w, dy, gamma, beta, mean, rstd, dact, dw, db,
-                         sum_dxhat, sum_dxhat_xhat, dgamma, dbeta, N, C, H, W,
-                         K);
-      break;
-    case 9:
-      hipLaunchKernelGGL((npf_cb2d_bwd_dact<9>), grid, blk, smem, stream, x,
-                         w, dy, gamma, beta, mean, rstd, dact, dw, db,
-                         sum_dxhat, sum_dxhat_xhat, dgamma, dbeta, N, C, H, W,
-                         K);
-      break;
-    case 11:
-      hipLaunchKernelGGL((npf_cb2d_bwd_dact<11>), grid, blk, smem, stream, x,
-                         w, dy, gamma, beta, mean, rstd, dact, dw, db,
-                         sum_dxhat, sum_dxhat_xhat, dgamma, dbeta, N, C, H, W,
-                         K);
-      break;
-    default:
-      hipLaunchKernelGGL((npf_cb2d_bwd_dact<0>), grid, blk, smem, stream, x,
-                         w, dy, gamma, beta, mean, rstd, dact, dw, db,
-                         sum_dxhat, sum_dxhat_xhat, dgamma, dbeta, N, C, H, W,
-                         K);
+    case 5: CB2D_BWD(5); break;
+    case 9: CB2D_BWD(9); break;
+    case 11: CB2D_BWD(11); break;
+    default: CB2D_BWD(0);
   }
+#undef CB2D_BWD
+}
+
+extern "C" void npf_cb2d_bwd_dact_launch(
+    const void* x, const float* w, const void* dy, const float* gamma,
+    const float* beta, const float* mean, const float* rstd, void* dact,
+    float* dw, float* db, float* sum_dxhat, float* sum_dxhat_xhat,
+    float* dgamma, float* dbeta, int N, int C, int H, int W, int K,
+    int is_bf16, hipStream_t stream) {
+  if (is_bf16)
+    cb2d_bwd_dispatch<__hip_bfloat16>(x, w, dy, gamma, beta, mean, rstd,
+                                      dact, dw, db, sum_dxhat,
+                                      sum_dxhat_xhat, dgamma, dbeta, N, C, H,
+                                      W, K, stream);
+  else
+    cb2d_bwd_dispatch<float>(x, w, dy, gamma, beta, mean, rstd, dact, dw, db,
+                             sum_dxhat, sum_dxhat_xhat, dgamma, dbeta, N, C,
+                             H, W, K, stream);
 }

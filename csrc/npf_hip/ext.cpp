@@ -52,7 +52,7 @@ void npf_gauss_ll_fwd_launch(const float*, const float*, const float*, float*,
 void npf_gauss_ll_bwd_launch(const float*, const float*, const float*,
                              const float*, float*, float*, long long,
                              long long, hipStream_t);
-void npf_cb_stats_launch(const float*, float*, float*, float*, float*,
+void npf_cb_stats_launch(const void*, int, float*, float*, float*, float*,
                          float*, float*, int, int, int, float, float,
                          hipStream_t);
 void npf_cb_fwd_launch(const float*, const float*, const float*, const float*,
@@ -62,19 +62,19 @@ void npf_cb_bwd_launch(const float*, const float*, const float*, const float*,
                        const float*, const float*, const float*, float*,
                        float*, float*, float*, float*, float*, float*, float*,
                        int, int, int, int, int, hipStream_t);
-void npf_cb_bwd_dx_launch(const float*, const float*, const float*,
+void npf_cb_bwd_dx_launch(const void*, const void*, const void*,
                           const float*, const float*, const float*,
-                          const float*, const float*, float*, int, int, int,
-                          int, hipStream_t);
-void npf_cb2d_fwd_launch(const float*, const float*, const float*,
+                          const float*, const float*, void*, int, int, int,
+                          int, int, hipStream_t);
+void npf_cb2d_fwd_launch(const void*, const void*, const float*,
                          const float*, const float*, const float*,
-                         const float*, const float*, float*, int, int, int,
-                         int, int, hipStream_t);
-void npf_cb2d_bwd_dact_launch(const float*, const float*, const float*,
+                         const float*, const float*, void*, int, int, int,
+                         int, int, int, hipStream_t);
+void npf_cb2d_bwd_dact_launch(const void*, const float*, const void*,
                               const float*, const float*, const float*,
-                              const float*, float*, float*, float*, float*,
-                              float*, float*, float*, int, int, int, int, int,
-                              hipStream_t);
+                              const float*, void*, float*, float*, float*,
+                              float*, float*, float*, int, int, int, int,
+                              int, int, hipStream_t);
 void npf_gde_fwd_launch(const float*, const float*, const float*, float*, int,
                         int, int, int, int, hipStream_t);
 void npf_gde_bwd_launch(const float*, const float*, const float*, const float*,
@@ -267,19 +267,19 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> convblock_stats(
     torch::Tensor x, torch::Tensor running_mean, torch::Tensor running_var,
     double eps, double momentum) {
   check_cuda_contig(x, "x");
-  TORCH_CHECK(x.scalar_type() == torch::kFloat32, "convblock is fp32");
   const int N = x.size(0), C = x.size(1), L = x.size(2);
-  auto mean = torch::empty({C}, x.options());
-  auto rstd = torch::empty({C}, x.options());
-  auto save_var = torch::empty({C}, x.options());
-  auto sums = torch::zeros({2 * C}, x.options());  // (sum ; sumsq) workspace
+  auto fopts = x.options().dtype(torch::kFloat32);
+  auto mean = torch::empty({C}, fopts);
+  auto rstd = torch::empty({C}, fopts);
+  auto save_var = torch::empty({C}, fopts);
+  auto sums = torch::zeros({2 * C}, fopts);  // (sum ; sumsq) workspace
   auto stream = at::hip::getCurrentHIPStream();
   float* rm = running_mean.defined() ? running_mean.data_ptr<float>() : nullptr;
   float* rv = running_var.defined() ? running_var.data_ptr<float>() : nullptr;
-  npf_cb_stats_launch(x.data_ptr<float>(), sums.data_ptr<float>(),
-                      mean.data_ptr<float>(), rstd.data_ptr<float>(),
-                      save_var.data_ptr<float>(), rm, rv, N, C, L, (float)eps,
-                      (float)momentum, stream);
+  npf_cb_stats_launch(x.data_ptr(), is_bf16(x) ? 1 : 0,
+                      sums.data_ptr<float>(), mean.data_ptr<float>(),
+                      rstd.data_ptr<float>(), save_var.data_ptr<float>(), rm,
+                      rv, N, C, L, (float)eps, (float)momentum, stream);
   return {mean, rstd, save_var};
 }
 
@@ -352,20 +352,19 @@ torch::Tensor convblock2d_fwd(torch::Tensor x, torch::Tensor res,
                               torch::Tensor mean, torch::Tensor rstd) {
   check_cuda_contig(x, "x");
   check_cuda_contig(w, "w");
-  TORCH_CHECK(x.scalar_type() == torch::kFloat32, "convblock is fp32");
   const int N = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
   const int K = w.size(-1);
   TORCH_CHECK(K <= 13 && K % 2 == 1, "2D kernel size must be odd and <= 13");
   auto y = torch::empty_like(x);
   auto stream = at::hip::getCurrentHIPStream();
   npf_cb2d_fwd_launch(
-      x.data_ptr<float>(), res.defined() ? res.data_ptr<float>() : nullptr,
+      x.data_ptr(), res.defined() ? res.data_ptr() : nullptr,
       w.data_ptr<float>(), b.defined() ? b.data_ptr<float>() : nullptr,
       gamma.defined() ? gamma.data_ptr<float>() : nullptr,
       gamma.defined() ? beta.data_ptr<float>() : nullptr,
       gamma.defined() ? mean.data_ptr<float>() : nullptr,
-      gamma.defined() ? rstd.data_ptr<float>() : nullptr, y.data_ptr<float>(),
-      N, C, H, W, K, stream);
+      gamma.defined() ? rstd.data_ptr<float>() : nullptr, y.data_ptr(), N, C,
+      H, W, K, is_bf16(x) ? 1 : 0, stream);
   return y;
 }
 
@@ -382,35 +381,36 @@ std::vector<torch::Tensor> convblock2d_bwd(torch::Tensor x, torch::Tensor w,
   const int N = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
   const int K = w.size(-1);
   const bool has_bn = gamma.defined();
+  const int bf = is_bf16(x) ? 1 : 0;
   auto dact = torch::empty_like(x);
   auto dx = torch::empty_like(x);
   auto dw = torch::zeros_like(w);
-  auto db = has_bias ? torch::zeros({C}, x.options()) : torch::Tensor();
-  auto opts = x.options();
+  auto opts = x.options().dtype(torch::kFloat32);
+  auto db = has_bias ? torch::zeros({C}, opts) : torch::Tensor();
   auto sum_dxhat = has_bn ? torch::zeros({C}, opts) : torch::Tensor();
   auto sum_dxhat_xhat = has_bn ? torch::zeros({C}, opts) : torch::Tensor();
   auto dgamma = has_bn ? torch::zeros({C}, opts) : torch::Tensor();
   auto dbeta = has_bn ? torch::zeros({C}, opts) : torch::Tensor();
   auto stream = at::hip::getCurrentHIPStream();
   npf_cb2d_bwd_dact_launch(
-      x.data_ptr<float>(), w.data_ptr<float>(), dy.data_ptr<float>(),
+      x.data_ptr(), w.data_ptr<float>(), dy.data_ptr(),
       has_bn ? gamma.data_ptr<float>() : nullptr,
       has_bn ? beta.data_ptr<float>() : nullptr,
       has_bn ? mean.data_ptr<float>() : nullptr,
-      has_bn ? rstd.data_ptr<float>() : nullptr, dact.data_ptr<float>(),
+      has_bn ? rstd.data_ptr<float>() : nullptr, dact.data_ptr(),
       dw.data_ptr<float>(), has_bias ? db.data_ptr<float>() : nullptr,
       has_bn ? sum_dxhat.data_ptr<float>() : nullptr,
       has_bn ? sum_dxhat_xhat.data_ptr<float>() : nullptr,
       has_bn ? dgamma.data_ptr<float>() : nullptr,
-      has_bn ? dbeta.data_ptr<float>() : nullptr, N, C, H, W, K, stream);
+      has_bn ? dbeta.data_ptr<float>() : nullptr, N, C, H, W, K, bf, stream);
   npf_cb_bwd_dx_launch(
-      x.data_ptr<float>(), dact.data_ptr<float>(), dy.data_ptr<float>(),
+      x.data_ptr(), dact.data_ptr(), dy.data_ptr(),
       has_bn ? gamma.data_ptr<float>() : nullptr,
       has_bn ? mean.data_ptr<float>() : nullptr,
       has_bn ? rstd.data_ptr<float>() : nullptr,
       has_bn ? sum_dxhat.data_ptr<float>() : nullptr,
-      has_bn ? sum_dxhat_xhat.data_ptr<float>() : nullptr,
-      dx.data_ptr<float>(), N, C, H * W, training ? 1 : 0, stream);
+      has_bn ? sum_dxhat_xhat.data_ptr<float>() : nullptr, dx.data_ptr(), N,
+      C, H * W, training ? 1 : 0, bf, stream);
   return {dx, dw, db, dgamma, dbeta};
 }
 

@@ -502,3 +502,38 @@ class TestHIPKernels:
             )
             assert dw_rel < 0.06, (i, float(dw_rel))
             assert db_rel < 0.06, (i, float(db_rel))
+
+    def test_conv_block_2d_bf16_io(self):
+        """bf16 I/O path (autocast regime): fp32-accumulated kernel vs the
+        bf16-rounded composed oracle."""
+        import torch.nn as nn
+
+        from npf.ops.functional import _conv_block2d_ref
+
+        g = torch.Generator(device="cuda").manual_seed(0)
+        N, C, H, W, K = 8, 64, 32, 32, 9
+        x = torch.randn(N, C, H, W, device="cuda", generator=g,
+                        dtype=torch.bfloat16, requires_grad=True)
+        conv = nn.Conv2d(C, C, K, padding=K // 2, groups=C).cuda()
+        norm = nn.BatchNorm2d(C).cuda().train()
+        res = torch.randn_like(x, requires_grad=True)
+
+        out = F_ops.conv_block_2d(x, conv, bn=norm, residual=res)
+        assert out.dtype == torch.bfloat16
+
+        conv0 = nn.Conv2d(C, C, K, padding=K // 2, groups=C)
+        conv0.load_state_dict({k: v.cpu() for k, v in conv.state_dict().items()})
+        norm0 = nn.BatchNorm2d(C)
+        norm0.train()
+        x0 = x.detach().cpu().float().requires_grad_(True)
+        r0 = res.detach().cpu().float().requires_grad_(True)
+        ref = _conv_block2d_ref(x0, conv0.weight, conv0.bias, norm0, r0, True)
+        assert torch.allclose(out.cpu().float(), ref, atol=0.15), (
+            (out.cpu().float() - ref).abs().max()
+        )
+        dout = torch.randn_like(out)
+        out.backward(dout)
+        ref.backward(dout.cpu().float())
+        assert torch.allclose(x.grad.cpu().float(), x0.grad, atol=0.2), (
+            (x.grad.cpu().float() - x0.grad).abs().max()
+        )
