@@ -173,6 +173,11 @@ class NPFTrainer:
         self.history = []
         self._best_valid = float("inf")
 
+    @property
+    def module_(self):
+        """skorch-compatible alias (reference code reads `trainer.module_`)."""
+        return self.module
+
     # ------------------------------------------------------------------ #
     # loops
     # ------------------------------------------------------------------ #

@@ -122,3 +122,74 @@ def test_sharded_eval_preserves_row_count_and_agreement():
     ll0, ll1 = out[0], out[1]
     assert ll0.shape == (24,)
     assert np.allclose(ll0, ll1)  # every rank reassembles the same full vector
+
+
+def _replica_sync_worker(rank, port, out):
+    """3 optimizer steps at world=2: replicas must stay bitwise identical."""
+    _init(rank, WORLD, port)
+    from npf.parallel import FlatDDP
+
+    set_seed(0)
+    model = CNP(1, 1, r_dim=16)
+    ddp = FlatDDP(model)
+    opt = torch.optim.Adam(model.parameters(), lr=1e-3)
+    crit = CNPFLoss()
+    crit.train()
+    model.train()
+    for step in range(3):
+        Xc, Yc, Xt, Yt = _make_batch(seed=1000 * step + rank)
+        ddp.zero_grad_()
+        loss = crit(model(Xc, Yc, Xt, Yt), Yt)
+        loss.backward()
+        ddp.reduce_()
+        opt.step()
+    out[rank] = torch.cat([p.detach().reshape(-1) for p in model.parameters()])
+    dist.destroy_process_group()
+
+
+def test_replicas_stay_identical_across_steps():
+    port = _free_port()
+    man = mp.Manager()
+    out = man.dict()
+    mp.spawn(_replica_sync_worker, args=(port, out), nprocs=WORLD, join=True)
+    assert torch.equal(out[0], out[1])
+
+
+def _ckpt_worker(rank, port, tmpdir, out):
+    """Checkpoint written under world=2 training must equal what a world-1
+    load sees (DP-invariant format, rank-0-only writes)."""
+    _init(rank, WORLD, port)
+    from npf.data import GPDataset, cntxt_trgt_collate
+    from npf.data.kernels import RBF
+    from npf.train import NPFTrainer
+    from npf.utils.datasplit import CntxtTrgtGetter, GetRandomIndcs
+
+    set_seed(0)
+    ds = GPDataset(kernel=RBF(0.2), n_samples=32, n_points=16)
+    collate = cntxt_trgt_collate(
+        CntxtTrgtGetter(contexts_getter=GetRandomIndcs(a=3, b=6))
+    )
+    trainer = NPFTrainer(
+        CNP(1, 1, r_dim=16), CNPFLoss(), collate_fn=collate, device="cpu",
+        batch_size=8, max_epochs=1, chckpnt_dirname=tmpdir, monitor=None,
+        seed=7,
+    )
+    trainer.fit(ds)
+    out[rank] = torch.cat(
+        [p.detach().reshape(-1) for p in trainer.module_.parameters()]
+    )
+    dist.destroy_process_group()
+
+
+def test_checkpoint_dp_invariant(tmp_path):
+    port = _free_port()
+    man = mp.Manager()
+    out = man.dict()
+    mp.spawn(
+        _ckpt_worker, args=(port, str(tmp_path), out), nprocs=WORLD, join=True
+    )
+    # replicas ended identical and the rank-0 checkpoint matches them
+    assert torch.equal(out[0], out[1])
+    sd = torch.load(os.path.join(str(tmp_path), "params.pt"), map_location="cpu")
+    flat = torch.cat([v.reshape(-1) for v in sd.values()])
+    assert torch.equal(flat, out[0])
