@@ -17,6 +17,10 @@
 
 #define CB2_BLOCK 256
 #define CB2_MAX_K 13
+#define CB2_TR 8   // plane rows per workgroup: small LDS tiles -> enough
+                   // resident waves to hide LDS latency (a full 64x64 plane
+                   // per WG was 41.5 KB -> 3 waves/SIMD -> 8.7ms bwd; row
+                   // tiles re-read the k/2 halo rows but run latency-hidden)
 
 template <typename T, int KT>
 __global__ void __launch_bounds__(CB2_BLOCK)
@@ -28,12 +32,13 @@ npf_cb2d_fwd(const T* __restrict__ x, const T* __restrict__ res,
   const int K = (KT > 0) ? KT : K_rt;
   extern __shared__ float smem[];
   const int pad = K / 2;
-  const int HP = H + 2 * pad, WP = W + 2 * pad;
-  float* a = smem;            // [HP * WP] activation plane with halo
-  float* ws = smem + HP * WP; // [K * K]
+  const int TRP = CB2_TR + 2 * pad, WP = W + 2 * pad;
+  float* a = smem;             // [TRP * WP] activation row-tile with halo
+  float* ws = smem + TRP * WP; // [K * K]
 
   const int n = blockIdx.x / C;
   const int c = blockIdx.x % C;
+  const int r0 = blockIdx.y * CB2_TR;
   const T* xpl = x + ((long)n * C + c) * H * W;
 
   const bool has_bn = gamma != nullptr;
@@ -42,8 +47,8 @@ npf_cb2d_fwd(const T* __restrict__ x, const T* __restrict__ res,
   const float gshift = has_bn ? beta[c] : 0.f;
 
   for (int i = threadIdx.x; i < K * K; i += CB2_BLOCK) ws[i] = w[c * K * K + i];
-  for (int i = threadIdx.x; i < HP * WP; i += CB2_BLOCK) {
-    const int r = i / WP - pad, col = i % WP - pad;
+  for (int i = threadIdx.x; i < TRP * WP; i += CB2_BLOCK) {
+    const int r = r0 + i / WP - pad, col = i % WP - pad;
     float v = 0.f;
     if (r >= 0 && r < H && col >= 0 && col < W)
       v = fmaxf((ldf(xpl + r * W + col) - mu) * gscale + gshift, 0.f);
@@ -51,11 +56,12 @@ npf_cb2d_fwd(const T* __restrict__ x, const T* __restrict__ res,
   }
   __syncthreads();
 
+  const int tr = min(CB2_TR, H - r0);
   const float b = (bias != nullptr) ? bias[c] : 0.f;
   T* ypl = y + ((long)n * C + c) * H * W;
   const T* rpl =
       (res != nullptr) ? res + ((long)n * C + c) * H * W : nullptr;
-  for (int i = threadIdx.x; i < H * W; i += CB2_BLOCK) {
+  for (int i = threadIdx.x; i < tr * W; i += CB2_BLOCK) {
     const int r = i / W, col = i % W;
     float acc = b;
     #pragma unroll
@@ -64,8 +70,9 @@ npf_cb2d_fwd(const T* __restrict__ x, const T* __restrict__ res,
       #pragma unroll
       for (int kc = 0; kc < K; ++kc) acc += ws[kr * K + kc] * arow[kc];
     }
-    if (rpl != nullptr) acc += ldf(rpl + i);
-    stf(ypl + i, acc);
+    const long gp = (long)(r0 + r) * W + col;
+    if (rpl != nullptr) acc += ldf(rpl + gp);
+    stf(ypl + gp, acc);
   }
 }
 
@@ -85,14 +92,15 @@ npf_cb2d_bwd_dact(const T* __restrict__ x, const float* __restrict__ w,
   const int K = (KT > 0) ? KT : K_rt;
   extern __shared__ float smem[];
   const int pad = K / 2;
-  const int HP = H + 2 * pad, WP = W + 2 * pad;
-  float* a = smem;                     // [HP*WP] activations (halo)
-  float* dys = smem + HP * WP;         // [HP*WP] dY (halo)
-  float* ws = dys + HP * WP;           // [K*K]
+  const int TRP = CB2_TR + 2 * pad, WP = W + 2 * pad;
+  float* a = smem;                      // [TRP*WP] activations (halo)
+  float* dys = smem + TRP * WP;         // [TRP*WP] dY (halo)
+  float* ws = dys + TRP * WP;           // [K*K]
   __shared__ float red[16];
 
   const int n = blockIdx.x / C;
   const int c = blockIdx.x % C;
+  const int r0 = blockIdx.y * CB2_TR;
   const T* xpl = x + ((long)n * C + c) * H * W;
   const T* dypl = dy + ((long)n * C + c) * H * W;
 
@@ -104,8 +112,8 @@ npf_cb2d_bwd_dact(const T* __restrict__ x, const float* __restrict__ w,
   const float gshift = has_bn ? beta[c] : 0.f;
 
   for (int i = threadIdx.x; i < K * K; i += CB2_BLOCK) ws[i] = w[c * K * K + i];
-  for (int i = threadIdx.x; i < HP * WP; i += CB2_BLOCK) {
-    const int r = i / WP - pad, col = i % WP - pad;
+  for (int i = threadIdx.x; i < TRP * WP; i += CB2_BLOCK) {
+    const int r = r0 + i / WP - pad, col = i % WP - pad;
     const bool in = (r >= 0 && r < H && col >= 0 && col < W);
     dys[i] = in ? ldf(dypl + r * W + col) : 0.f;
     float v = 0.f;
@@ -114,11 +122,13 @@ npf_cb2d_bwd_dact(const T* __restrict__ x, const float* __restrict__ w,
   }
   __syncthreads();
 
+  const int tr = min(CB2_TR, H - r0);
   T* dactpl = dact + ((long)n * C + c) * H * W;
   float s_dxhat = 0.f, s_dxhat_xhat = 0.f, s_dg = 0.f, s_db = 0.f, dbp = 0.f;
-  for (int i = threadIdx.x; i < H * W; i += CB2_BLOCK) {
+  for (int i = threadIdx.x; i < tr * W; i += CB2_BLOCK) {
     const int r = i / W, col = i % W;
-    const float dyl = ldf(dypl + i);
+    const long gp = (long)(r0 + r) * W + col;
+    const float dyl = ldf(dypl + gp);
     dbp += dyl;
     // transposed conv: da[p] = sum_k w[k] dY[p + pad - k]
     float da = 0.f;
@@ -130,9 +140,9 @@ npf_cb2d_bwd_dact(const T* __restrict__ x, const float* __restrict__ w,
     }
     const float act = a[(r + pad) * WP + col + pad];
     const float dr = (act > 0.f) ? da : 0.f;
-    stf(dactpl + i, dr);
+    stf(dactpl + gp, dr);
     if (has_bn) {
-      const float xhat = (ldf(xpl + i) - mu) * rs;
+      const float xhat = (ldf(xpl + gp) - mu) * rs;
       const float dxh = dr * gm;
       s_dxhat += dxh;
       s_dxhat_xhat += dxh * xhat;
@@ -141,10 +151,8 @@ npf_cb2d_bwd_dact(const T* __restrict__ x, const float* __restrict__ w,
     }
   }
   __syncthreads();
-  // dW[kr,kc] = sum_p dY[p] * a[p + (kr,kc)]: the K^2 weight positions are
-  // split across the 4 waves; each wave scans the whole plane from LDS and
-  // reduces with shuffles only — no block barriers (the barrier-per-position
-  // version measured 5.5ms/call at [256,128,64,64] k=9)
+  // dW[kr,kc] = sum_p dY[p] * a[p + (kr,kc)]: K^2 positions split across
+  // waves, shuffle-only reduction, one atomic per (row-tile, position)
   {
     const int lane = threadIdx.x & 63;
     const int wv = threadIdx.x >> 6;
@@ -152,7 +160,7 @@ npf_cb2d_bwd_dact(const T* __restrict__ x, const float* __restrict__ w,
     for (int kk = wv; kk < K * K; kk += nw) {
       const int kr = kk / K, kc = kk % K;
       float psum = 0.f;
-      for (int i = lane; i < H * W; i += 64) {
+      for (int i = lane; i < tr * W; i += 64) {
         const int r = i / W, col = i % W;
         psum += dys[(r + pad) * WP + col + pad] * a[(r + kr) * WP + col + kc];
       }
@@ -194,8 +202,9 @@ static void cb2d_fwd_dispatch(const void* x, const void* res, const float* w,
                               int W, int K, hipStream_t stream) {
   const int pad = K / 2;
   const size_t smem =
-      ((size_t)(H + 2 * pad) * (W + 2 * pad) + K * K) * sizeof(float);
-  const dim3 grid((unsigned)N * C), blk(CB2_BLOCK);
+      ((size_t)(CB2_TR + 2 * pad) * (W + 2 * pad) + K * K) * sizeof(float);
+  const dim3 grid((unsigned)N * C, (unsigned)((H + CB2_TR - 1) / CB2_TR)),
+      blk(CB2_BLOCK);
 #define CB2D_FWD(KT)                                                        \
   hipLaunchKernelGGL((npf_cb2d_fwd<T, KT>), grid, blk, smem, stream,        \
                      (const T*)x, (const T*)res, w, bias, gamma, beta,      \
@@ -233,8 +242,9 @@ static void cb2d_bwd_dispatch(const void* x, const float* w, const void* dy,
                               int H, int W, int K, hipStream_t stream) {
   const int pad = K / 2;
   const size_t smem =
-      (2 * (size_t)(H + 2 * pad) * (W + 2 * pad) + K * K) * sizeof(float);
-  const dim3 grid((unsigned)N * C), blk(CB2_BLOCK);
+      (2 * (size_t)(CB2_TR + 2 * pad) * (W + 2 * pad) + K * K) * sizeof(float);
+  const dim3 grid((unsigned)N * C, (unsigned)((H + CB2_TR - 1) / CB2_TR)),
+      blk(CB2_BLOCK);
 #define CB2D_BWD(KT)                                                        \
   hipLaunchKernelGGL((npf_cb2d_bwd_dact<T, KT>), grid, blk, smem, stream,   \
                      (const T*)x, w, (const T*)dy, gamma, beta, mean, rstd, \

@@ -454,11 +454,12 @@ class _ConvBlock2dFn(torch.autograd.Function):
 
 
 def _cb2d_lds_ok(x, k):
-    """The 2D backward stages two haloed planes in LDS; stay under the 64 KB
-    default dynamic-LDS limit, else fall back to the composed path."""
-    H, W = x.shape[-2], x.shape[-1]
+    """The 2D kernels stage 8-row tiles (+halo) in LDS; stay under the 64 KB
+    default dynamic-LDS limit, else fall back to the composed path (only
+    reachable for images wider than ~780 px)."""
+    W = x.shape[-1]
     pad = k // 2
-    return 2 * (H + 2 * pad) * (W + 2 * pad) * 4 <= 64 * 1024
+    return 2 * (8 + 2 * pad) * (W + 2 * pad) * 4 <= 64 * 1024
 
 
 def conv_block_2d(x, conv, bn=None, residual=None):
