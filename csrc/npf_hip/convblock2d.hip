@@ -61,18 +61,26 @@ npf_cb2d_fwd(const T* __restrict__ x, const T* __restrict__ res,
   T* ypl = y + ((long)n * C + c) * H * W;
   const T* rpl =
       (res != nullptr) ? res + ((long)n * C + c) * H * W : nullptr;
-  for (int i = threadIdx.x; i < tr * W; i += CB2_BLOCK) {
-    const int r = i / W, col = i % W;
-    float acc = b;
-    #pragma unroll
-    for (int kr = 0; kr < K; ++kr) {
-      const float* arow = a + (r + kr) * WP + col;
+  // div-free pixel walk: software integer div/mod per element was ~8x the
+  // useful VALU work (SQ_INSTS_VALU 5075/wave vs ~650 needed)
+  {
+    int r = threadIdx.x / W, col = threadIdx.x % W;
+    const int dr = CB2_BLOCK / W, dc = CB2_BLOCK % W;
+    while (r < tr) {
+      float acc = b;
       #pragma unroll
-      for (int kc = 0; kc < K; ++kc) acc += ws[kr * K + kc] * arow[kc];
+      for (int kr = 0; kr < K; ++kr) {
+        const float* arow = a + (r + kr) * WP + col;
+        #pragma unroll
+        for (int kc = 0; kc < K; ++kc) acc += ws[kr * K + kc] * arow[kc];
+      }
+      const long gp = (long)(r0 + r) * W + col;
+      if (rpl != nullptr) acc += ldf(rpl + gp);
+      stf(ypl + gp, acc);
+      r += dr;
+      col += dc;
+      if (col >= W) { col -= W; ++r; }
     }
-    const long gp = (long)(r0 + r) * W + col;
-    if (rpl != nullptr) acc += ldf(rpl + gp);
-    stf(ypl + gp, acc);
   }
 }
 
@@ -125,29 +133,36 @@ npf_cb2d_bwd_dact(const T* __restrict__ x, const float* __restrict__ w,
   const int tr = min(CB2_TR, H - r0);
   T* dactpl = dact + ((long)n * C + c) * H * W;
   float s_dxhat = 0.f, s_dxhat_xhat = 0.f, s_dg = 0.f, s_db = 0.f, dbp = 0.f;
-  for (int i = threadIdx.x; i < tr * W; i += CB2_BLOCK) {
-    const int r = i / W, col = i % W;
-    const long gp = (long)(r0 + r) * W + col;
-    const float dyl = ldf(dypl + gp);
-    dbp += dyl;
-    // transposed conv: da[p] = sum_k w[k] dY[p + pad - k]
-    float da = 0.f;
-    #pragma unroll
-    for (int kr = 0; kr < K; ++kr) {
-      const float* drow = dys + (r + K - 1 - kr) * WP + col;
+  {
+    int r = threadIdx.x / W, col = threadIdx.x % W;
+    const int drr = CB2_BLOCK / W, dcc = CB2_BLOCK % W;
+    while (r < tr) {
+      const long gp = (long)(r0 + r) * W + col;
+      const float dyl = ldf(dypl + gp);
+      dbp += dyl;
+      // transposed conv: da[p] = sum_k w[k] dY[p + pad - k]
+      float da = 0.f;
       #pragma unroll
-      for (int kc = 0; kc < K; ++kc) da += ws[kr * K + kc] * drow[K - 1 - kc];
-    }
-    const float act = a[(r + pad) * WP + col + pad];
-    const float dr = (act > 0.f) ? da : 0.f;
-    stf(dactpl + gp, dr);
-    if (has_bn) {
-      const float xhat = (ldf(xpl + gp) - mu) * rs;
-      const float dxh = dr * gm;
-      s_dxhat += dxh;
-      s_dxhat_xhat += dxh * xhat;
-      s_dg += dr * xhat;
-      s_db += dr;
+      for (int kr = 0; kr < K; ++kr) {
+        const float* drow = dys + (r + K - 1 - kr) * WP + col;
+        #pragma unroll
+        for (int kc = 0; kc < K; ++kc)
+          da += ws[kr * K + kc] * drow[K - 1 - kc];
+      }
+      const float act = a[(r + pad) * WP + col + pad];
+      const float dr = (act > 0.f) ? da : 0.f;
+      stf(dactpl + gp, dr);
+      if (has_bn) {
+        const float xhat = (ldf(xpl + gp) - mu) * rs;
+        const float dxh = dr * gm;
+        s_dxhat += dxh;
+        s_dxhat_xhat += dxh * xhat;
+        s_dg += dr * xhat;
+        s_db += dr;
+      }
+      r += drr;
+      col += dcc;
+      if (col >= W) { col -= W; ++r; }
     }
   }
   __syncthreads();
@@ -157,12 +172,17 @@ npf_cb2d_bwd_dact(const T* __restrict__ x, const float* __restrict__ w,
     const int lane = threadIdx.x & 63;
     const int wv = threadIdx.x >> 6;
     const int nw = CB2_BLOCK / 64;
+    const int r_in = lane / W, c_in = lane % W;
+    const int drw = 64 / W, dcw = 64 % W;
     for (int kk = wv; kk < K * K; kk += nw) {
       const int kr = kk / K, kc = kk % K;
       float psum = 0.f;
-      for (int i = lane; i < tr * W; i += 64) {
-        const int r = i / W, col = i % W;
+      int r = r_in, col = c_in;
+      while (r < tr) {
         psum += dys[(r + pad) * WP + col + pad] * a[(r + kr) * WP + col + kc];
+        r += drw;
+        col += dcw;
+        if (col >= W) { col -= W; ++r; }
       }
       psum = wave_reduce_sum(psum);
       if (lane == 0) atomicAdd(&dw[c * K * K + kk], psum);
