@@ -407,8 +407,8 @@ def plot_img_marginal_pred(
             predictive_all.base_dist.scale[:, i : i + 1, ...],
         )
         out = marginal_log_like(predictive, arange_marg).detach().reshape(1000, -1).numpy()
-        sarles = sarle(out)
-        if np.median(sarles) < best:
+        sarles = np.nan_to_num(sarle(out), nan=np.inf)
+        if i == 0 or np.median(sarles) < best:
             best = float(np.median(sarles))
             best_out, best_sarles = out, sarles
             best_pred = predictive
@@ -490,8 +490,8 @@ def plot_qualitative_with_kde(
 
     if percentiles is None:
         percentiles = np.linspace(0, 100, n_images)
-    values = np.percentile(test_loglike, percentiles, interpolation="nearest")
-    img_indcs = [int(np.argwhere(test_loglike == v)[0]) for v in values]
+    values = np.percentile(test_loglike, percentiles, method="nearest")
+    img_indcs = [int(np.argwhere(test_loglike == v)[0, 0]) for v in values]
     if is_smallest_xrange:
         axes[0].set_xlim(values[0] - 1, values[-1] + 1)
     for v in values:

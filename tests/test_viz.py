@@ -192,3 +192,41 @@ def test_plot_posterior_samples_grid_model(img_ds):
         is_uniform_grid=True, n_plots=2, is_return=True,
     )
     assert grid.dim() == 3 and grid.size(0) == 3
+
+
+def test_plot_img_marginal_pred(img_ds):
+    from model_zoo import attnlnp_2d
+    from npf.utils.datasplit import GridCntxtTrgtGetter, RandomMasker
+    from npf.viz import plot_img_marginal_pred
+
+    model = attnlnp_2d(y_dim=3)
+    model.n_z_samples_test = 3
+    getter = GridCntxtTrgtGetter(context_masker=RandomMasker(a=0.05, b=0.2))
+    fig = plot_img_marginal_pred(
+        model, img_ds, getter, n_samples=2, is_uniform_grid=False,
+        n_plots_loop=1, n_marginals=3,
+    )
+    assert len(fig.axes) == 2
+
+
+def test_plot_qualitative_with_kde(tmp_path, img_ds):
+    import numpy as np
+
+    from model_zoo import gridconvcnp_2d
+    from npf import CNPFLoss
+    from npf.train import NPFTrainer
+    from npf.viz import plot_qualitative_with_kde
+
+    trainer = NPFTrainer(
+        gridconvcnp_2d(y_dim=3), CNPFLoss(), device="cpu",
+        chckpnt_dirname=str(tmp_path),
+    )
+    rng = np.random.RandomState(0)
+    np.savetxt(
+        os.path.join(str(tmp_path), "eval.csv"),
+        rng.randn(len(img_ds)) * 50 + 300,
+    )
+    fig = plot_qualitative_with_kde(
+        ["GridConvCNP", trainer], img_ds, n_images=2, n_samples=1,
+    )
+    assert len(fig.axes) == 2
