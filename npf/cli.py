@@ -78,7 +78,8 @@ def _build_data(args):
             from npf.recipes import get_datasets_variable_kernel_gp
 
             d, t, v = get_datasets_variable_kernel_gp(
-                n_samples=args.n_tasks, save_file=args.data_cache
+                n_samples=args.n_tasks, save_file=args.data_cache,
+                n_test=args.n_test_tasks,
             )
         else:
             d, t, v = get_gp_datasets(
@@ -88,6 +89,7 @@ def _build_data(args):
                 n_points=128,
                 is_vary_kernel_hyp=args.data == "Variable_Matern_Kernel",
                 is_reuse_across_epochs=False,
+                n_test=args.n_test_tasks,
             )
         return d, t, v, True
     train, test = get_img_datasets([args.data])
@@ -192,6 +194,7 @@ def main(argv=None):
         q.add_argument("--seed", type=int, default=123)
         q.add_argument("--runs", type=int, default=1)
         q.add_argument("--n-tasks", type=int, default=50000)
+        q.add_argument("--n-test-tasks", type=int, default=10000)
         q.add_argument("--chckpnt-dir", default="results/")
         q.add_argument("--data-cache", default="data/gp_dataset.npz")
         q.add_argument("--device", default=None)

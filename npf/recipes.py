@@ -68,7 +68,7 @@ def get_img_datasets(datasets):
     return train_datasets, test_datasets
 
 
-def get_datasets_single_gp(**kwargs):
+def get_datasets_single_gp(n_test=10000, **kwargs):
     """The three fixed-hyperparameter GP benchmarks (reference
     ntbks_helpers.py:78-99: RBF ls=0.2; periodic ls=0.5 p=0.5; noisy Matern
     White(0.1)+Matern ls=0.2 nu=1.5; 50k tasks x 128 points, fresh every
@@ -86,7 +86,7 @@ def get_datasets_single_gp(**kwargs):
         is_reuse_across_epochs=False,
     )
     defaults.update(kwargs)
-    return get_gp_datasets(kernels, **defaults)
+    return get_gp_datasets(kernels, n_test=n_test, **defaults)
 
 
 def get_datasets_variable_hyp_gp(**kwargs):
@@ -106,9 +106,11 @@ def get_datasets_variable_hyp_gp(**kwargs):
     return get_gp_datasets(kernels, **defaults)
 
 
-def get_datasets_variable_kernel_gp(**kwargs):
+def get_datasets_variable_kernel_gp(n_test=10000, **kwargs):
     """All single-GP datasets merged into one task distribution."""
-    datasets, test_datasets, valid_datasets = get_datasets_single_gp(**kwargs)
+    datasets, test_datasets, valid_datasets = get_datasets_single_gp(
+        n_test=n_test, **kwargs
+    )
     return (
         dict(All_Kernels=DatasetMerger(datasets.values())),
         dict(All_Kernels=DatasetMerger(test_datasets.values())),
@@ -138,10 +140,11 @@ def sample_gp_dataset_like(dataset, **kwargs):
     return new_dataset
 
 
-def get_gp_datasets(kernels, save_file="data/gp_dataset.npz", **kwargs):
+def get_gp_datasets(kernels, save_file="data/gp_dataset.npz", n_test=10000,
+                    **kwargs):
     """(train, test, valid) GPDataset dicts for each named kernel; test is a
-    10k-task fixed draw (cache chunk -1), valid n/10 (chunk -2) — reference
-    ntbks_helpers.py:136-172."""
+    fixed draw of `n_test` tasks (cache chunk -1, 10k in the reference),
+    valid n/10 (chunk -2) — reference ntbks_helpers.py:136-172."""
 
     def chunk_file(name):
         return (save_file, name) if save_file is not None else None
@@ -152,7 +155,7 @@ def get_gp_datasets(kernels, save_file="data/gp_dataset.npz", **kwargs):
     }
     datasets_test = {
         k: sample_gp_dataset_like(
-            ds, save_file=chunk_file(k), idx_chunk=-1, n_samples=10000
+            ds, save_file=chunk_file(k), idx_chunk=-1, n_samples=n_test
         )
         for k, ds in datasets.items()
     }
