@@ -118,3 +118,20 @@ def test_load_params_roundtrip(tmp_path):
     t2.load_params()
     for p1, p2 in zip(t1.module.parameters(), t2.module.parameters()):
         assert torch.equal(p1, p2)
+
+
+def test_cli_train_eval_roundtrip(tmp_path):
+    """End-to-end CLI: tiny train writes the reference file set; eval reloads."""
+    from npf import cli
+
+    cli.main([
+        "train", "--model", "CNP", "--data", "RBF_Kernel",
+        "--epochs", "1", "--n-tasks", "64", "--n-test-tasks", "64",
+        "--batch-size", "32",
+        "--chckpnt-dir", str(tmp_path) + "/",
+        "--data-cache", str(tmp_path / "gp.npz"),
+        "--device", "cpu",
+    ])
+    run = tmp_path / "RBF_Kernel" / "CNP" / "run_0"
+    for f in ("params.pt", "optimizer.pt", "model_summary.txt", "eval.csv"):
+        assert (run / f).exists(), f
