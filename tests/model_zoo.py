@@ -69,6 +69,16 @@ def lnp_1d(r_dim=R_DIM):
     )
 
 
+def lnp_2d(y_dim=3, r_dim=R_DIM):
+    return LNP(
+        x_dim=2, y_dim=y_dim, is_q_zCct=True, n_z_samples_train=1, n_z_samples_test=32,
+        XYEncoder=merge_flat_input(
+            partial(MLP, n_hidden_layers=2, hidden_size=r_dim * 3), is_sum_merge=True
+        ),
+        **mlp_kwargs(r_dim),
+    )
+
+
 def attncnp_1d(r_dim=R_DIM):
     return AttnCNP(
         x_dim=1, y_dim=1,
@@ -158,6 +168,7 @@ PUBLISHED_PARAM_COUNTS = {
     "cnp_1d": 252098,
     "cnp_2d": 367750,
     "lnp_1d": 301634,
+    "lnp_2d": 417286,
     "attncnp_1d": 252738,
     "attncnp_2d": 386054,
     "attnlnp_1d": 335170,
@@ -172,6 +183,7 @@ BUILDERS = {
     "cnp_1d": cnp_1d,
     "cnp_2d": cnp_2d,
     "lnp_1d": lnp_1d,
+    "lnp_2d": lnp_2d,
     "attncnp_1d": attncnp_1d,
     "attncnp_2d": attncnp_2d,
     "attnlnp_1d": attnlnp_1d,

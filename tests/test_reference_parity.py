@@ -167,3 +167,94 @@ def test_eval_loglike_matches_published_mean():
     # published 149.16 on 10k tasks; ~128 count draws put the mean inside
     # +-60 nats (3 sigma) unless the model/data pipeline is wrong
     assert 89 < mean < 209, mean
+
+
+PRETRAINED_2D = os.path.join(REF, "results", "pretrained", "celeba32")
+
+REF_MODEL_DIRS_2D = {
+    "cnp_2d": "CNP",
+    "lnp_2d": "LNP",
+    "attncnp_2d": "AttnCNP",
+    "attnlnp_2d": "AttnLNP",
+    "gridconvcnp_2d": "ConvCNP",
+    "gridconvlnp_2d": "ConvLNP",
+}
+
+
+@needs_ref
+@pytest.mark.parametrize("name", sorted(REF_MODEL_DIRS_2D))
+def test_checkpoint_key_space_identical_2d(name):
+    sd_ref = torch.load(
+        os.path.join(PRETRAINED_2D, REF_MODEL_DIRS_2D[name], "run_0", "params.pt"),
+        map_location="cpu",
+    )
+    model = BUILDERS[name](y_dim=3)
+    sd_mine = model.state_dict()
+    assert set(sd_ref) == set(sd_mine)
+    for k in sd_ref:
+        assert sd_ref[k].shape == sd_mine[k].shape, k
+    model.load_state_dict(sd_ref)
+
+
+_REF_RUNNER_2D = r"""
+import sys, warnings, torch
+warnings.filterwarnings("ignore")
+sys.path.insert(0, "/root/reference")
+from functools import partial
+import torch.nn as nn
+from npf import GridConvCNP
+from npf.architectures import CNN, MLP, ResConvBlock, discard_ith_arg
+
+inp_path, out_path = sys.argv[1], sys.argv[2]
+d = torch.load(inp_path, weights_only=False)
+
+model = GridConvCNP(
+    x_dim=1, y_dim=3,
+    CNN=partial(CNN, ConvBlock=ResConvBlock, Conv=nn.Conv2d,
+                Normalization=nn.BatchNorm2d, n_blocks=5, kernel_size=9,
+                is_chan_last=True, n_conv_layers=2),
+    r_dim=128,
+    Decoder=discard_ith_arg(partial(MLP, n_hidden_layers=4, hidden_size=128), i=0),
+)
+sd = torch.load(
+    "/root/reference/results/pretrained/celeba32/ConvCNP/run_0/params.pt",
+    map_location="cpu",
+)
+model.load_state_dict(sd)
+model.eval()
+with torch.no_grad():
+    p, *_ = model(d["mc"], d["Y"], d["mt"])
+torch.save((p.base_dist.loc, p.base_dist.scale), out_path)
+"""
+
+
+@needs_ref
+def test_gridconvcnp_forward_matches_reference(tmp_path):
+    """Shipped celeba32 GridConvCNP weights + identical mask batch through
+    the reference implementation and ours."""
+    g = torch.Generator().manual_seed(7)
+    B, H, W = 2, 32, 32
+    Y = torch.rand(B, H, W, 3, generator=g)
+    mc = torch.rand(B, H, W, 1, generator=g) < 0.3
+    mt = torch.ones(B, H, W, 1, dtype=torch.bool)
+    inp, outp = str(tmp_path / "i.pt"), str(tmp_path / "o.pt")
+    torch.save({"mc": mc, "Y": Y, "mt": mt}, inp)
+
+    env = dict(os.environ, PYTHONPATH=REF)
+    subprocess.run(
+        [sys.executable, "-c", _REF_RUNNER_2D, inp, outp],
+        check=True, cwd="/tmp", env=env, capture_output=True,
+    )
+    rl, rs = torch.load(outp, weights_only=False)
+
+    m = BUILDERS["gridconvcnp_2d"](y_dim=3)
+    sd = torch.load(
+        os.path.join(PRETRAINED_2D, "ConvCNP", "run_0", "params.pt"),
+        map_location="cpu",
+    )
+    m.load_state_dict(sd)
+    m.eval()
+    with torch.no_grad():
+        p, *_ = m(mc, Y, mt)
+    assert (p.base_dist.loc - rl).abs().max() <= 1e-5
+    assert (p.base_dist.scale - rs).abs().max() <= 1e-5
