@@ -46,6 +46,7 @@ __all__ = [
     "StrFormatter",
     "PRETTY_RENAMER",
     "add_y_dim",
+    "get_std_processing_kwargs",
     "get_n_cntxt",
     "plot_multi_posterior_samples_1d",
     "plot_multi_posterior_samples_imgs",
@@ -221,6 +222,23 @@ PRETTY_RENAMER = StrFormatter(
     },
     to_upper=["Mnist", "Svhn", "Cnp", "Lnp", "Rbf"],
 )
+
+
+def get_std_processing_kwargs(min_sigma_pred=0.01, min_lat=None):
+    """Predictive/latent std lower-bound kwargs — the knobs of the 24-model
+    loss-ablation grid (reference Losses.ipynb `get_std_processing_kwargs`;
+    results tabulated in BASELINE.md)."""
+    import torch.nn.functional as F
+
+    kwargs = dict(
+        p_y_scale_transformer=lambda y_scale: min_sigma_pred
+        + (1 - min_sigma_pred) * F.softplus(y_scale)
+    )
+    if min_lat is not None:
+        kwargs["q_z_scale_transformer"] = lambda z_scale: min_lat + (
+            1 - min_lat
+        ) * F.softplus(z_scale)
+    return kwargs
 
 
 def add_y_dim(models, datasets):

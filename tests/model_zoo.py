@@ -15,6 +15,7 @@ from npf import (
     GridConvCNP,
     GridConvLNP,
 )
+from npf.utils.helpers import CircularPad2d, make_abs_conv, make_padded_conv
 from npf.architectures import (
     CNN,
     MLP,
@@ -152,6 +153,30 @@ def convlnp_1d(r_dim=R_DIM):
     )
 
 
+def gridconvcnp_zsmms(y_dim=1, r_dim=R_DIM):
+    """Fully translation-equivariant zsmms variant: circular padding in every
+    conv incl. the density encoder (reference ConvCNP.ipynb model_2d_extrap;
+    pretrained at results/pretrained/zsmms/ConvCNP)."""
+    return GridConvCNP(
+        x_dim=1, y_dim=y_dim,
+        CNN=partial(
+            CNN, Conv=make_padded_conv(torch.nn.Conv2d, CircularPad2d),
+            Normalization=partial(torch.nn.BatchNorm2d, eps=1e-2),
+            n_blocks=5, kernel_size=9, **CNN_KWARGS,
+        ),
+        Conv=lambda y_dim: make_padded_conv(
+            make_abs_conv(torch.nn.Conv2d), CircularPad2d
+        )(y_dim, y_dim, groups=y_dim, kernel_size=11, padding=11 // 2, bias=False),
+        r_dim=r_dim,
+        Decoder=discard_ith_arg(partial(MLP, n_hidden_layers=4, hidden_size=r_dim), i=0),
+    )
+
+
+def gridconvcnp_xl(y_dim=3, r_dim=R_DIM):
+    """12-block celeba128 XL config (reference ConvCNP.ipynb model_2d_XL)."""
+    return gridconvcnp_2d(y_dim=y_dim, r_dim=r_dim, n_blocks=12)
+
+
 def gridconvlnp_2d(y_dim=3, r_dim=R_DIM):
     return GridConvLNP(
         x_dim=1, y_dim=y_dim,
@@ -175,6 +200,7 @@ PUBLISHED_PARAM_COUNTS = {
     "attnlnp_2d": 468486,
     "convcnp_1d": 276612,
     "gridconvcnp_2d": 340721,
+    "gridconvcnp_xl": 722417,
     "convlnp_1d": 376068,
     "gridconvlnp_2d": 487793,
 }
@@ -190,6 +216,8 @@ BUILDERS = {
     "attnlnp_2d": attnlnp_2d,
     "convcnp_1d": convcnp_1d,
     "gridconvcnp_2d": gridconvcnp_2d,
+    "gridconvcnp_zsmms": gridconvcnp_zsmms,
+    "gridconvcnp_xl": gridconvcnp_xl,
     "convlnp_1d": convlnp_1d,
     "gridconvlnp_2d": gridconvlnp_2d,
 }

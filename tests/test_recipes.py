@@ -126,3 +126,35 @@ def test_gp_dataset_generator_oracle(tiny_gp):
     gen.fit([[0.0], [0.5]], [[0.1], [0.2]])
     mean, std = gen.predict([[0.25]], return_std=True)
     assert mean.shape[0] == 1
+
+
+def test_std_processing_kwargs_ablation_grid():
+    """The LatLB/SigLB ablation knobs build working LNP variants."""
+    from model_zoo import R_DIM
+    from functools import partial
+    from npf import LNP
+    from npf.architectures import MLP, merge_flat_input
+
+    for min_lat in (None, 0.1):
+        for min_sig in (0.01, 1e-4):
+            kw = recipes.get_std_processing_kwargs(
+                min_sigma_pred=min_sig, min_lat=min_lat
+            )
+            m = LNP(
+                x_dim=1, y_dim=1, is_q_zCct=False,
+                n_z_samples_train=2, n_z_samples_test=2,
+                XYEncoder=merge_flat_input(
+                    partial(MLP, n_hidden_layers=2, hidden_size=2 * R_DIM),
+                    is_sum_merge=True,
+                ),
+                XEncoder=partial(MLP, n_hidden_layers=1, hidden_size=R_DIM),
+                Decoder=merge_flat_input(
+                    partial(MLP, n_hidden_layers=4, hidden_size=R_DIM),
+                    is_sum_merge=True,
+                ),
+                r_dim=R_DIM,
+                **kw,
+            )
+            X = torch.rand(2, 9, 1) * 2 - 1
+            p, *_ = m(X[:, :4], torch.randn(2, 4, 1), X)
+            assert float(p.base_dist.scale.min()) >= min_sig

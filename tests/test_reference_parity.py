@@ -258,3 +258,35 @@ def test_gridconvcnp_forward_matches_reference(tmp_path):
         p, *_ = m(mc, Y, mt)
     assert (p.base_dist.loc - rl).abs().max() <= 1e-5
     assert (p.base_dist.scale - rs).abs().max() <= 1e-5
+
+
+@needs_ref
+def test_zsmms_circularpad_checkpoint_parity():
+    """Fully-translation-equivariant zsmms variant (CircularPad2d in every
+    conv) matches the shipped zsmms/ConvCNP checkpoint key-for-key."""
+    sd = torch.load(
+        os.path.join(REF, "results", "pretrained", "zsmms", "ConvCNP",
+                     "run_0", "params.pt"),
+        map_location="cpu",
+    )
+    import model_zoo as zoo
+
+    m = zoo.gridconvcnp_zsmms(y_dim=1)
+    assert set(sd) == set(m.state_dict())
+    m.load_state_dict(sd)
+    # forward on a 56x56 zsmms-shaped batch
+    g = torch.Generator().manual_seed(0)
+    Y = torch.rand(2, 56, 56, 1, generator=g)
+    mc = torch.rand(2, 56, 56, 1, generator=g) < 0.2
+    mt = torch.ones(2, 56, 56, 1, dtype=torch.bool)
+    m.eval()
+    with torch.no_grad():
+        p, *_ = m(mc, Y, mt)
+    assert torch.isfinite(p.base_dist.loc).all()
+
+
+def test_gridconvcnp_xl_param_count():
+    import model_zoo as zoo
+
+    m = zoo.gridconvcnp_xl(y_dim=3)
+    assert sum(p.numel() for p in m.parameters()) == 722417
