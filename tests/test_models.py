@@ -134,3 +134,56 @@ def test_state_dict_roundtrip():
     for (n1, p1), (n2, p2) in zip(m1.named_parameters(), m2.named_parameters()):
         assert n1 == n2
         assert torch.equal(p1, p2)
+
+
+def test_gridconvcnp_zsmms_circular_translation_equivariance():
+    """The CircularPad2d variant is EXACTLY equivariant to circular shifts:
+    roll(inputs) -> roll(predictions) (the property zsmms training relies
+    on, reference ConvCNP.ipynb 'full translation equivariance')."""
+    import sys
+
+    sys.path.insert(0, "tests")
+    from model_zoo import gridconvcnp_zsmms
+
+    torch.manual_seed(0)
+    m = gridconvcnp_zsmms(y_dim=1).eval()
+    g = torch.Generator().manual_seed(1)
+    Y = torch.rand(1, 16, 16, 1, generator=g)
+    mc = torch.rand(1, 16, 16, 1, generator=g) < 0.3
+    mt = torch.ones(1, 16, 16, 1, dtype=torch.bool)
+    with torch.no_grad():
+        p, *_ = m(mc, Y, mt)
+        loc = p.base_dist.loc[0, 0]  # [H, W, 1]
+        sh = (5, 3)
+        p2, *_ = m(torch.roll(mc, sh, dims=(1, 2)), torch.roll(Y, sh, dims=(1, 2)), mt)
+        loc2 = p2.base_dist.loc[0, 0]
+    assert torch.allclose(torch.roll(loc, sh, dims=(0, 1)), loc2, atol=1e-5), (
+        (torch.roll(loc, sh, dims=(0, 1)) - loc2).abs().max()
+    )
+
+
+def test_convcnp_translation_equivariance_1d():
+    """ConvCNP is (approximately, up to grid discretization) translation
+    equivariant: shifting context+target x by a grid-aligned offset shifts
+    predictions (the ConvCNP paper's core property)."""
+    import sys
+
+    sys.path.insert(0, "tests")
+    from model_zoo import convcnp_1d
+
+    torch.manual_seed(0)
+    m = convcnp_1d().eval()
+    g = torch.Generator().manual_seed(2)
+    Xc = torch.rand(1, 9, 1, generator=g) * 0.8 - 0.6  # keep room to shift
+    Yc = torch.randn(1, 9, 1, generator=g)
+    Xt = torch.linspace(-0.6, 0.2, 33).view(1, -1, 1)
+    # shift by an exact multiple of the induced-grid spacing
+    delta = 16 / (m.density_induced * 3 - 1) * (3.0)  # grid step * 16... use steps
+    step = 3.0 / (int(m.density_induced * 3) - 1)
+    delta = 24 * step
+    with torch.no_grad():
+        p1, *_ = m(Xc, Yc, Xt)
+        p2, *_ = m(Xc + delta, Yc, Xt + delta)
+    a = p1.base_dist.loc[0, 0, :, 0]
+    b = p2.base_dist.loc[0, 0, :, 0]
+    assert torch.allclose(a, b, atol=1e-3), (a - b).abs().max()
