@@ -167,3 +167,51 @@ def test_synthetic_images():
     assert cls.shape == (3, 32, 32)
     tr, te = get_train_test_img_dataset("synthetic32")
     assert len(tr) > 0 and len(te) > 0
+
+
+class TestSplitterExtras:
+    def test_superresolution_getter(self):
+        from npf.utils.datasplit import SuperresolutionCntxtTrgtGetter
+
+        getter = SuperresolutionCntxtTrgtGetter(resolution_factor=1 / 4)
+        X = torch.rand(2, 3, 16, 16)
+        Xc, Yc, Xt, Yt = getter(X)
+        # context = the subsampled grid of the low-res(-upsampled) image:
+        # (16/4)^2 = 16 points; target = the full original image
+        assert Yc.shape == (2, 16, 3)
+        assert Yt.shape == (2, 16 * 16, 3)
+        assert torch.allclose(
+            Yt.view(2, 16, 16, 3), X.permute(0, 2, 3, 1)
+        )
+
+    def test_half_maskers(self):
+        from npf.utils.datasplit import half_masker
+
+        m = half_masker(2, (8, 8), dim=0)
+        assert m.shape == (2, 8, 8, 1)
+        assert m[:, :4].all() and not m[:, 4:].any()
+        m1 = half_masker(2, (8, 8), dim=1)
+        assert m1[:, :, :4].all() and not m1[:, :, 4:].any()
+
+    def test_resolution_masker(self):
+        from npf.utils.datasplit import ResolutionMasker
+
+        m = ResolutionMasker(factor=4)(2, (8, 8))
+        assert m.sum() == 2 * 4  # (8/4)^2 per image (centered offset 2)
+        assert m[0, 2, 2] and m[0, 2, 6] and m[0, 6, 2]
+
+    def test_mask_combinators(self):
+        from npf.utils.datasplit import and_masks, half_masker, not_masks, or_masks
+
+        a = half_masker(1, (4, 4), dim=0)
+        b = half_masker(1, (4, 4), dim=1)
+        assert (and_masks(a, b) == (a & b)).all()
+        assert (or_masks(a, b) == (a | b)).all()
+        assert (not_masks(a, b) == (a & ~b)).all()
+
+    def test_beta_binomial_counts(self):
+        from npf.utils.datasplit import GetRandomIndcs
+
+        g = GetRandomIndcs(a=3.0, b=20.0, is_beta_binomial=True)
+        idcs = g(4, 64)
+        assert 0 <= idcs.shape[1] <= 64
