@@ -187,3 +187,32 @@ def test_convcnp_translation_equivariance_1d():
     a = p1.base_dist.loc[0, 0, :, 0]
     b = p2.base_dist.loc[0, 0, :, 0]
     assert torch.allclose(a, b, atol=1e-3), (a - b).abs().max()
+
+
+def test_convcnp_with_unet_and_forced_bottleneck():
+    """UNet induced-to-induced CNN incl. the batch-half bottleneck averaging
+    (reference cnn.py:466-475 is_force_same_bottleneck) runs fwd+bwd."""
+    from functools import partial
+
+    from npf import CNP, ConvCNP  # noqa: F401
+    from npf.architectures import MLP, ResConvBlock, SetConv, UnetCNN, discard_ith_arg
+
+    m = ConvCNP(
+        x_dim=1, y_dim=1, Interpolator=SetConv,
+        CNN=partial(
+            UnetCNN, ConvBlock=ResConvBlock, Conv=torch.nn.Conv1d,
+            Normalization=torch.nn.Identity, n_blocks=3, kernel_size=5,
+            is_chan_last=True, n_conv_layers=1, Pool=torch.nn.AvgPool1d,
+            upsample_mode="linear", is_force_same_bottleneck=True,
+        ),
+        density_induced=16, r_dim=32,
+        Decoder=discard_ith_arg(partial(MLP, n_hidden_layers=2, hidden_size=32), i=0),
+    )
+    m.train()
+    # duplicated batch: two context draws of the same functions
+    Xc = torch.rand(4, 6, 1) * 2 - 1
+    Yc = torch.randn(4, 6, 1)
+    Xt = torch.rand(4, 12, 1) * 2 - 1
+    p, *_ = m(Xc, Yc, Xt)
+    p.base_dist.loc.sum().backward()
+    assert p.base_dist.loc.shape == (1, 4, 12, 1)
