@@ -633,6 +633,9 @@ def mlp_chain(x, weights, biases):
     autocast semantics), weights fp32 [out, in] per layer, out bf16.
     """
     lead = x.shape[:-1]
+    d_out = weights[-1].shape[0]
+    if x.numel() == 0:  # zero-context episodes: no rows, no launch
+        return x.new_zeros(*lead, d_out, dtype=torch.bfloat16)
     xb = x.reshape(-1, x.shape[-1]).to(torch.bfloat16).contiguous()
     y = _MLPChainFn.apply(xb, len(weights), *weights, *biases)
     return y.reshape(*lead, y.shape[-1])
