@@ -150,6 +150,8 @@ class NPFTrainer:
         self.decay_lr = decay_lr
         self.patience = patience
         self.grad_clip_norm = grad_clip_norm
+        if isinstance(amp_dtype, str):  # "bfloat16" / "float16" from the CLI
+            amp_dtype = getattr(torch, amp_dtype)
         self.amp_dtype = amp_dtype
         self.num_workers = num_workers
         self.is_progressbar = is_progressbar
