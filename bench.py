@@ -285,6 +285,25 @@ def main():
         opt.step()
         return loss
 
+    def fwd_bwd_step():
+        """Capturable core only (no collective, no optimizer) — the partial
+        graph used when RCCL ops refuse capture at world > 1."""
+        ddp.zero_grad_()
+        if use_bf16:
+            with torch.autocast("cuda", dtype=torch.bfloat16):
+                out = model(
+                    X_cntxt=sXc, Y_cntxt=sYc, X_trgt=sXt,
+                    Y_trgt=sYt if pass_y else None,
+                )
+        else:
+            out = model(
+                X_cntxt=sXc, Y_cntxt=sYc, X_trgt=sXt,
+                Y_trgt=sYt if pass_y else None,
+            )
+        loss = crit(out, sYt.float())
+        loss.backward()
+        return loss
+
     def load(i):
         for buf, t in zip((sXc, sYc, sXt, sYt), pool[i % POOL_BATCHES]):
             buf.copy_(t)
@@ -296,39 +315,55 @@ def main():
     if use_cuda:
         torch.cuda.synchronize()
 
-    # ---- optional hipGraph capture of the whole train step ----
+    # ---- optional hipGraph capture: full step, else fwd+bwd only ----
     graph = None
+    graph_is_full = True
     if use_cuda and not args.no_graph and cfg.get("graph_ok", True):
         import gc
+
+        def try_capture(step_fn):
+            s = torch.cuda.Stream()
+            s.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(s):
+                for i in range(3):
+                    load(i)
+                    wl = step_fn()
+            del wl
+            gc.collect()
+            torch.cuda.current_stream().wait_stream(s)
+            torch.cuda.synchronize()
+            g = torch.cuda.CUDAGraph()
+            load(0)
+            with torch.cuda.graph(g):
+                sl = step_fn()
+            return g, sl
 
         # drop every reference to the warmup autograd graph: a live
         # AccumulateGrad from a pre-capture iteration breaks capture
         loss = float(loss.detach())
         gc.collect()
         try:
-            s = torch.cuda.Stream()
-            s.wait_stream(torch.cuda.current_stream())
-            with torch.cuda.stream(s):
-                for i in range(3):
-                    load(i)
-                    wl = train_step()
-            del wl
-            gc.collect()
-            torch.cuda.current_stream().wait_stream(s)
-            torch.cuda.synchronize()
-            graph = torch.cuda.CUDAGraph()
-            load(0)
-            with torch.cuda.graph(graph):
-                static_loss = train_step()
-            log("[bench] hipGraph capture OK")
-        except Exception as e:  # fall back to eager
-            log(f"[bench] graph capture failed ({e!r}); running eager")
+            graph, static_loss = try_capture(train_step)
+            log("[bench] hipGraph capture OK (full step)")
+        except Exception as e:
+            log(f"[bench] full-step capture failed ({e!r}); trying fwd+bwd-only")
             graph = None
+            gc.collect()
+            try:
+                graph, static_loss = try_capture(fwd_bwd_step)
+                graph_is_full = False
+                log("[bench] hipGraph capture OK (fwd+bwd; reduce+step eager)")
+            except Exception as e2:
+                log(f"[bench] graph capture failed ({e2!r}); running eager")
+                graph = None
 
     def timed_step(i):
         load(i)
         if graph is not None:
             graph.replay()
+            if not graph_is_full:
+                ddp.reduce_()
+                opt.step()
         else:
             train_step()
 
