@@ -216,3 +216,26 @@ def test_convcnp_with_unet_and_forced_bottleneck():
     p, *_ = m(Xc, Yc, Xt)
     p.base_dist.loc.sum().backward()
     assert p.base_dist.loc.shape == (1, 4, 12, 1)
+
+
+@pytest.mark.parametrize("name", ["cnp_1d", "attncnp_1d", "convcnp_1d"])
+def test_context_permutation_invariance(name):
+    """NP predictions are invariant to the ORDER of the context set."""
+    import sys
+
+    sys.path.insert(0, "tests")
+    import model_zoo as zoo
+
+    torch.manual_seed(0)
+    m = zoo.BUILDERS[name]().eval()
+    g = torch.Generator().manual_seed(3)
+    Xc = torch.rand(2, 11, 1, generator=g) * 2 - 1
+    Yc = torch.randn(2, 11, 1, generator=g)
+    Xt = torch.rand(2, 17, 1, generator=g) * 2 - 1
+    perm = torch.randperm(11, generator=g)
+    with torch.no_grad():
+        p1, *_ = m(Xc, Yc, Xt)
+        p2, *_ = m(Xc[:, perm], Yc[:, perm], Xt)
+    assert torch.allclose(
+        p1.base_dist.loc, p2.base_dist.loc, atol=1e-5
+    ), (p1.base_dist.loc - p2.base_dist.loc).abs().max()
