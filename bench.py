@@ -227,6 +227,10 @@ def main():
         # NPF_BENCH_NO_TUNE=1 skips it (profiling runs: tuning floods the
         # kernel stats with thousands of one-off candidate launches)
         try:
+            # per-rank results file: 8 ranks must not write the same csv
+            os.environ.setdefault(
+                "PYTORCH_TUNABLEOP_FILENAME", f"tunableop_results{rank}.csv"
+            )
             torch.cuda.tunable.enable(True)
             torch.cuda.tunable.tuning_enable(True)
         except Exception:
