@@ -98,13 +98,16 @@ class ConvCNP(NeuralProcessFamily):
         batch_size, n_cntxt, _ = X_cntxt.shape
         X_induced = self._get_X_induced(X_cntxt)
 
-        # context -> induced grid: [B, M, r_dim]
-        R_induced = self.cntxt_to_induced(X_cntxt, X_induced, Y_cntxt)
         if n_cntxt == 0:
-            # empty context: zero representation (density channel is zero too)
+            # empty context: zero representation (density channel is zero
+            # too) — skip the SetConv entirely rather than launching a
+            # degenerate K=0 kernel
             R_induced = torch.zeros(
-                batch_size, self.n_induced, self.r_dim, device=R_induced.device
+                batch_size, self.n_induced, self.r_dim, device=X_cntxt.device
             )
+        else:
+            # context -> induced grid: [B, M, r_dim]
+            R_induced = self.cntxt_to_induced(X_cntxt, X_induced, Y_cntxt)
 
         # induced -> induced: the CNN stack
         return self.induced_to_induced(R_induced)
