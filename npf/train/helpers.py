@@ -6,7 +6,6 @@ Capability parity with /root/reference/utils/helpers.py (load_all_results
 :77-93).
 """
 
-import contextlib
 import glob
 import logging
 import os

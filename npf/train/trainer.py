@@ -16,8 +16,6 @@ LR decay, seed handling — with an MI355X-first loop:
 import logging
 import os
 import time
-from copy import deepcopy
-
 import numpy as np
 import torch
 from torch.optim import Adam

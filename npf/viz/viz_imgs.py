@@ -9,7 +9,6 @@ import logging
 import math
 import os
 import random
-from functools import partial
 
 import matplotlib.pyplot as plt
 import matplotlib.ticker as ticker
