@@ -247,9 +247,32 @@ class MultiheadAttender(nn.Module):
 
     def forward(self, keys, queries, values, rel_pos_enc=None, **kwargs):
         """keys [B,K,kq], queries [B,Q,kq], values [B,K,v] -> [B,Q,out]."""
-        keys = self.key_transform(keys)
-        queries = self.query_transform(queries)
-        values = self.value_transform(values)
+        if (
+            keys is queries
+            and queries is values
+            and not self.is_relative_pos
+            and self.kq_size == self.value_size
+        ):
+            # self-attention: ONE GEMM for all three projections (the
+            # weights are concatenated; only the query has a bias)
+            w = torch.cat(
+                [
+                    self.key_transform.weight,
+                    self.query_transform.weight,
+                    self.value_transform.weight,
+                ],
+                dim=0,
+            )
+            proj = torch.nn.functional.linear(keys, w)
+            keys, queries, values = proj.split(
+                [self.kq_size, self.kq_size, self.value_size], dim=-1
+            )
+            if self.query_transform.bias is not None:
+                queries = queries + self.query_transform.bias
+        else:
+            keys = self.key_transform(keys)
+            queries = self.query_transform(queries)
+            values = self.value_transform(values)
 
         queries = self._make_multiheaded(queries, self.kq_head_size)
         values = self._make_multiheaded(values, self.value_head_size)

@@ -62,7 +62,12 @@ class SelfAttention(nn.Module):
 
         out = X
         for attn_layer in self.attn_layers:
-            out = attn_layer(out + add_to_keys, out, out)
+            # keep keys IS queries IS values when there is no positional
+            # offset: skips an add kernel and lets the attender fuse its
+            # three projections into one GEMM
+            keys = out if isinstance(add_to_keys, int) and add_to_keys == 0 \
+                else out + add_to_keys
+            out = attn_layer(keys, out, out)
 
         if self.is_resize:
             out = self.resize(out)
