@@ -135,3 +135,55 @@ def test_cli_train_eval_roundtrip(tmp_path):
     run = tmp_path / "RBF_Kernel" / "CNP" / "run_0"
     for f in ("params.pt", "optimizer.pt", "model_summary.txt", "eval.csv"):
         assert (run / f).exists(), f
+
+
+def test_load_all_results(tmp_path):
+    """Aggregates {data}/{model}/run_*/eval.csv into the reference DataFrame
+    schema (reference utils/helpers.py:22-32)."""
+    import numpy as np
+
+    from npf.train import load_all_results
+
+    for data, model, vals in [
+        ("RBF_Kernel", "CNP", [1.0, 3.0]),
+        ("RBF_Kernel", "ConvCNP", [10.0, 20.0]),
+    ]:
+        d = tmp_path / data / model / "run_0"
+        d.mkdir(parents=True)
+        np.savetxt(d / "eval.csv", np.array(vals))
+    df = load_all_results(str(tmp_path) + "/")
+    assert set(df.columns) == {"Data", "Model", "Runs", "LogLike"}
+    assert len(df) == 2
+    row = df[df.Model == "ConvCNP"].iloc[0]
+    assert row.LogLike == 15.0
+
+
+def test_sample_predictor():
+    import sys
+
+    sys.path.insert(0, "tests")
+    from model_zoo import cnp_1d
+    from npf.utils.predict import SamplePredictor
+
+    m = cnp_1d().eval()
+    Xc = torch.rand(2, 5, 1) * 2 - 1
+    Yc = torch.randn(2, 5, 1)
+    Xt = torch.rand(2, 9, 1) * 2 - 1
+    loc = SamplePredictor(m)(Xc, Yc, Xt)
+    assert loc.shape == (1, 2, 9, 1) and not loc.requires_grad
+    dist = SamplePredictor(m, is_dist=True)(Xc, Yc, Xt)
+    assert torch.allclose(dist.base_dist.loc, loc)
+
+
+def test_dataset_merger_attr_forwarding():
+    from npf.data import GPDataset
+    from npf.data.helpers import DatasetMerger
+    from npf.data.kernels import RBF
+
+    a = GPDataset(kernel=RBF(0.2), n_samples=4, n_points=8)
+    b = GPDataset(kernel=RBF(0.4), n_samples=4, n_points=8)
+    m = DatasetMerger([a, b])
+    assert len(m) == 8
+    assert m.min_max == a.min_max  # attr forwarded from the first dataset
+    x, y = m[5]
+    assert x.shape == (8, 1)
