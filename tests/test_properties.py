@@ -6,7 +6,6 @@ GPU kernels' oracles)."""
 
 import math
 
-import numpy as np
 import torch
 from hypothesis import given, settings
 from hypothesis import strategies as st
