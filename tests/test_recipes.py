@@ -157,4 +157,4 @@ def test_std_processing_kwargs_ablation_grid():
             )
             X = torch.rand(2, 9, 1) * 2 - 1
             p, *_ = m(X[:, :4], torch.randn(2, 4, 1), X)
-            assert float(p.base_dist.scale.min()) >= min_sig
+            assert float(p.base_dist.scale.detach().min()) >= min_sig
