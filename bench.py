@@ -39,7 +39,6 @@ import time
 import torch
 
 sys.path.insert(0, os.path.abspath(os.path.dirname(__file__)))
-sys.path.insert(0, os.path.join(os.path.abspath(os.path.dirname(__file__)), "tests"))
 
 LR = 1e-3
 POOL_BATCHES = 16  # pre-generated synthetic task pool (cycled)
@@ -144,7 +143,7 @@ def _img_grid_pool(device, batch, seed, shape=(3, 64, 64), cntxt_frac=0.1):
 def _configs():
     from npf import CNPFLoss, ELBOLossLNPF, NLLLossLNPF
     from npf.data.kernels import RBF, ExpSineSquared
-    import model_zoo as zoo
+    from npf import zoo
 
     return {
         "attncnp": dict(

@@ -7,10 +7,9 @@ import time
 import torch
 
 sys.path.insert(0, os.path.abspath(os.path.join(os.path.dirname(__file__), "..")))
-sys.path.insert(0, os.path.abspath(os.path.join(os.path.dirname(__file__), "..", "tests")))
 
 from npf import CNPFLoss
-from model_zoo import attncnp_1d
+from npf.zoo import attncnp_1d
 
 
 def mm_bwd_microbench():

@@ -1,11 +1,11 @@
 """attnlnp2d loss-component trace: find what goes non-finite."""
 import os, sys, torch
-sys.path.insert(0, "/root/repo"); sys.path.insert(0, "/root/repo/tests")
+sys.path.insert(0, "/root/repo")
 import bench
 from npf import ELBOLossLNPF
 from npf.losses import sum_log_prob
 from npf.ops import functional as F_ops
-import model_zoo as zoo
+from npf import zoo
 
 def run(steps=200, autocast=True, eager_ops=False):
     os.environ["NPF_FORCE_EAGER"] = "1" if eager_ops else "0"

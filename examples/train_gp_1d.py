@@ -15,9 +15,8 @@ from functools import partial
 from pathlib import Path
 
 sys.path.insert(0, str(Path(__file__).parent.parent))
-sys.path.insert(0, str(Path(__file__).parent.parent / "tests"))
 
-import model_zoo as zoo
+from npf import zoo
 from npf import CNPFLoss, ELBOLossLNPF, NLLLossLNPF
 from npf.data.dataloader import cntxt_trgt_collate
 from npf.recipes import get_datasets_single_gp

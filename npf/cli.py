@@ -2,7 +2,7 @@
 
 The reference has no CLI (SURVEY.md §5.6 — notebooks compose
 functools.partial trees); this maps the same canonical configurations
-(npf.recipes + the model zoo in bench_utils/model_zoo) onto flags:
+(npf.recipes + the model zoo in npf.zoo) onto flags:
 
     python -m npf.cli train --model ConvCNP --data RBF_Kernel \
         --epochs 10 --batch-size 32 --chckpnt-dir results/
@@ -22,7 +22,6 @@ import sys
 
 # the canonical model builders live next to the tests so they are shared by
 # tests, benches and this CLI without duplicating configs
-sys.path.insert(0, os.path.join(os.path.dirname(__file__), "..", "tests"))
 
 MODELS_1D = {
     "CNP": "cnp_1d",
@@ -124,7 +123,7 @@ def cmd_list(_args):
 
 
 def _common(args, is_retrain):
-    import model_zoo as zoo
+    from npf import zoo
     from functools import partial
     from npf.train import train_models
 
