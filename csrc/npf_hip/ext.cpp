@@ -458,6 +458,11 @@ torch::Tensor gauss_kl_fwd(torch::Tensor mq, torch::Tensor sq,
                            torch::Tensor mp, torch::Tensor sp) {
   for (auto* t : {&mq, &sq, &mp, &sp}) check_cuda_contig(*t, "kl input");
   TORCH_CHECK(mq.scalar_type() == torch::kFloat32, "gauss_kl is fp32");
+  // the kernel indexes all four tensors with mq's layout: reject
+  // broadcastable (size-1) dims instead of reading out of bounds
+  TORCH_CHECK(sq.sizes() == mq.sizes() && mp.sizes() == mq.sizes() &&
+                  sp.sizes() == mq.sizes(),
+              "gauss_kl requires equal shapes for mq/sq/mp/sp");
   const long long rows = mq.size(0);
   const long long m = mq.numel() / rows;
   auto out = torch::empty({rows}, mq.options());

@@ -129,18 +129,36 @@ class ResConvBlock(nn.Module):
     def reset_parameters(self):
         weights_init(self)
 
+    @staticmethod
+    def _is_plain_same_pad(conv, dims):
+        """True only for stock zero same-padding convs.  Subclasses that pad
+        externally (e.g. make_padded_conv's PaddedConv: native padding=0 plus
+        a CircularPad2d `padder`) must NOT take the fused path — the fused
+        kernel applies zero same-padding and would silently drop circular
+        padding (and with it the zsmms variant's exact shift equivariance)."""
+        k = conv.kernel_size
+        return (
+            not hasattr(conv, "padder")
+            and conv.padding_mode == "zeros"
+            and tuple(conv.padding) == tuple(ks // 2 for ks in k)
+        )
+
     def _fused_op(self, X):
         from npf.ops import conv_block_1d, conv_block_2d, has_extension
 
         if not (X.is_cuda and isinstance(self.activation, nn.ReLU)
                 and has_extension()):
             return None
+        convs = [self.conv2_depthwise]
+        if self.n_conv_layers == 2:
+            convs.append(self.conv1.depthwise)
         if (X.dim() == 3 and isinstance(self.conv2_depthwise, nn.Conv1d)
-                and isinstance(self.norm2, (nn.BatchNorm1d, nn.Identity))):
+                and isinstance(self.norm2, (nn.BatchNorm1d, nn.Identity))
+                and all(self._is_plain_same_pad(c, 1) for c in convs)):
             return conv_block_1d
         if (X.dim() == 4 and isinstance(self.conv2_depthwise, nn.Conv2d)
                 and isinstance(self.norm2, (nn.BatchNorm2d, nn.Identity))
-                and self.conv2_depthwise.padding_mode == "zeros"):
+                and all(self._is_plain_same_pad(c, 2) for c in convs)):
             return conv_block_2d
         return None
 

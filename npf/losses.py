@@ -103,6 +103,11 @@ class ELBOLossLNPF(BaseLossNPF):
             and isinstance(q_zCct.base_dist, Normal)
             and isinstance(q_zCc, Independent)
             and isinstance(q_zCc.base_dist, Normal)
+            # the fused kernel indexes by q_zCct's layout: all four tensors
+            # must have identical (non-broadcast) shapes
+            and q_zCct.base_dist.loc.shape == q_zCc.base_dist.loc.shape
+            and q_zCct.base_dist.scale.shape == q_zCct.base_dist.loc.shape
+            and q_zCc.base_dist.scale.shape == q_zCc.base_dist.loc.shape
         ):
             # fused HIP kernel: analytic KL + latent-set reduce in one pass
             E_z_kl = ops.gaussian_kl_sum(

@@ -340,6 +340,14 @@ class _ConvBlock1dFn(torch.autograd.Function):
         )
 
 
+def _cb1d_lds_ok(x, k):
+    """The 1D backward stages 2*CB_TN(=4) padded rows of fp32 in dynamic LDS
+    (convblock.hip npf_cb_bwd_launch): stay under the 64 KB limit, else fall
+    back to the composed path (only reachable for L over ~2030)."""
+    L = x.shape[-1]
+    return 2 * 4 * (L + k - 1) * 4 <= 64 * 1024
+
+
 def conv_block_1d(x, conv, bn=None, residual=None):
     """Fused norm->relu->depthwise-conv1d(+residual) on [N, C, L].
 
@@ -347,7 +355,11 @@ def conv_block_1d(x, conv, bn=None, residual=None):
     nn.BatchNorm1d (its running stats are updated in training mode exactly
     like torch), `residual` an optional tensor added to the output.
     """
-    if not x.is_cuda or _backend.require_extension("conv_block_1d") is None:
+    if (
+        not x.is_cuda
+        or not _cb1d_lds_ok(x, conv.weight.shape[-1])
+        or _backend.require_extension("conv_block_1d") is None
+    ):
         return _conv_block_ref(
             x, conv.weight, conv.bias, bn,
             residual, bn.training if bn is not None else conv.training,
@@ -532,11 +544,24 @@ class _GridDensityFn(torch.autograd.Function):
         return dx, None, dw.view(ctx.kshape)
 
 
+def _gde_lds_ok(x, k):
+    """griddensity.hip stages 2 padded 8-row tiles + the KxK filter in
+    dynamic LDS; stay under the 64 KB limit, else compose (only reachable
+    for images wider than ~780 px)."""
+    W = x.shape[-1]
+    pad = k // 2
+    return (2 * (8 + 2 * pad) * (W + 2 * pad) + k * k) * 4 <= 64 * 1024
+
+
 def grid_density(x, mask, weight):
     """Fused density encoder: x [B,C,H,W], mask [B,C or 1,H,W] (no grad),
     abs-conv weight [C,1,K,K] -> [B, 2C, H, W] (normalized signal ; density).
     """
-    if not x.is_cuda or _backend.require_extension("grid_density") is None:
+    if (
+        not x.is_cuda
+        or not _gde_lds_ok(x, weight.shape[-1])
+        or _backend.require_extension("grid_density") is None
+    ):
         return _grid_density_ref(x, mask, weight)
     mask = mask.float().expand_as(x).contiguous()
     return _GridDensityFn.apply(
@@ -634,8 +659,16 @@ def mlp_chain(x, weights, biases):
     """
     lead = x.shape[:-1]
     d_out = weights[-1].shape[0]
-    if x.numel() == 0:  # zero-context episodes: no rows, no launch
-        return x.new_zeros(*lead, d_out, dtype=torch.bfloat16)
+    if x.numel() == 0:
+        # zero-context episodes: no rows, no launch — but keep the output
+        # connected to every parameter so .grad is zero-filled (not absent):
+        # a rank whose whole batch is empty must still produce gradients for
+        # the DDP flat all-reduce to stay aligned across ranks
+        zero = x.reshape(-1).sum()
+        for t in list(weights) + list(biases):
+            zero = zero + t.sum() * 0.0
+        out = x.new_zeros(*lead, d_out, dtype=torch.bfloat16)
+        return out + zero.to(torch.bfloat16) * 0.0
     xb = x.reshape(-1, x.shape[-1]).to(torch.bfloat16).contiguous()
     y = _MLPChainFn.apply(xb, len(weights), *weights, *biases)
     return y.reshape(*lead, y.shape[-1])

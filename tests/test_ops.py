@@ -537,3 +537,78 @@ class TestHIPKernels:
         assert torch.allclose(x.grad.cpu().float(), x0.grad, atol=0.2), (
             (x.grad.cpu().float() - x0.grad).abs().max()
         )
+
+
+class TestAdviceGuards:
+    """Round-2 guards from ADVICE.md (dispatch/LDS/zero-row edge cases)."""
+
+    def test_padded_conv_rejected_by_fused_dispatch(self):
+        # the zsmms CircularPad variant's PaddedConv (native padding=0 +
+        # separate circular padder) must not pass the fused-kernel guard:
+        # the fused path would zero same-pad and drop circular padding
+        from npf.architectures.cnn import ResConvBlock
+        from npf.utils.helpers import CircularPad2d, make_padded_conv
+
+        Conv = make_padded_conv(torch.nn.Conv2d, CircularPad2d)
+        blk = ResConvBlock(8, 8, Conv, kernel_size=5, n_conv_layers=2)
+        assert not blk._is_plain_same_pad(blk.conv2_depthwise, 2)
+        assert not blk._is_plain_same_pad(blk.conv1.depthwise, 2)
+        # a plain same-padded conv passes
+        blk2 = ResConvBlock(8, 8, torch.nn.Conv2d, kernel_size=5)
+        assert blk2._is_plain_same_pad(blk2.conv2_depthwise, 2)
+
+    def test_lds_guards(self):
+        from npf.ops.functional import _cb1d_lds_ok, _cb2d_lds_ok, _gde_lds_ok
+
+        ok = torch.zeros(1, 8, 128)
+        big = torch.zeros(1, 8, 4096)
+        assert _cb1d_lds_ok(ok, 19) and not _cb1d_lds_ok(big, 19)
+        ok2 = torch.zeros(1, 8, 32, 32)
+        big2 = torch.zeros(1, 8, 32, 2048)
+        assert _cb2d_lds_ok(ok2, 9) and not _cb2d_lds_ok(big2, 9)
+        assert _gde_lds_ok(ok2, 11) and not _gde_lds_ok(big2, 11)
+
+    def test_mlp_chain_zero_rows_still_flows_grads(self):
+        # an all-empty batch must still produce (zero) grads for every
+        # parameter, or DDP flat-buffer reduction desynchronizes ranks
+        from npf.ops.functional import mlp_chain
+
+        ws = [torch.randn(16, 8, requires_grad=True),
+              torch.randn(4, 16, requires_grad=True)]
+        bs = [torch.randn(16, requires_grad=True),
+              torch.randn(4, requires_grad=True)]
+        x = torch.zeros(0, 8, requires_grad=True)
+        out = mlp_chain(x, ws, bs)
+        assert out.shape == (0, 4)
+        out.sum().backward()
+        for t in ws + bs:
+            assert t.grad is not None
+            assert torch.all(t.grad == 0)
+
+
+@pytest.mark.gpu
+class TestZsmmsFusedParity:
+    def test_zsmms_model_gpu_matches_cpu(self):
+        """The circular-padded zsmms GridConvCNP must produce the same
+        forward on GPU (where fused dispatch is reachable) as the CPU
+        composed path — i.e. the fused kernels must NOT engage for
+        PaddedConv blocks (ADVICE.md high finding)."""
+        from npf.zoo import gridconvcnp_zsmms
+
+        torch.manual_seed(0)
+        m = gridconvcnp_zsmms(y_dim=1)
+        m0 = gridconvcnp_zsmms(y_dim=1)
+        m0.load_state_dict(m.state_dict())
+        m = m.cuda().eval()
+        m0.eval()
+
+        B, H, W = 2, 24, 24
+        g = torch.Generator().manual_seed(3)
+        mask_c = torch.rand(B, H, W, 1, generator=g) < 0.3
+        mask_t = torch.ones(B, H, W, 1, dtype=torch.bool)
+        Y = torch.rand(B, H, W, 1, generator=g) * 2 - 1
+        out0 = m0(mask_c, Y, mask_t)[0].base_dist.loc
+        out = m(mask_c.cuda(), Y.cuda(), mask_t.cuda())[0].base_dist.loc
+        assert torch.allclose(out.cpu(), out0, atol=5e-4), (
+            (out.cpu() - out0).abs().max()
+        )
