@@ -25,7 +25,6 @@ def cntxt_trgt_collate(get_cntxt_trgt, is_duplicate_batch=False, **kwargs):
             X = torch.cat([X, X], dim=0)
             if y is not None:
                 y = torch.cat([y, y], dim=0)
-            y = torch.cat([y, y], dim=0)
 
         X_cntxt, Y_cntxt, X_trgt, Y_trgt = get_cntxt_trgt(X, y, **kwargs)
         inputs = dict(X_cntxt=X_cntxt, Y_cntxt=Y_cntxt, X_trgt=X_trgt, Y_trgt=Y_trgt)

@@ -215,3 +215,108 @@ class TestSplitterExtras:
         g = GetRandomIndcs(a=3.0, b=20.0, is_beta_binomial=True)
         idcs = g(4, 64)
         assert 0 <= idcs.shape[1] <= 64
+
+
+class TestSplitterRewrite:
+    """Round-2 splitter: torch-native, device-capable, stratified counts."""
+
+    def test_uniform_count_marginal(self):
+        torch.manual_seed(0)
+        g = GetRandomIndcs(a=2, b=6)
+        counts = [g(4, 32).shape[1] for _ in range(400)]
+        assert min(counts) == 2 and max(counts) == 6
+        # roughly uniform over {2..6}
+        for c in range(2, 7):
+            frac = sum(1 for x in counts if x == c) / len(counts)
+            assert 0.1 < frac < 0.3, (c, frac)
+
+    def test_rows_are_distinct_subsets(self):
+        torch.manual_seed(0)
+        g = GetRandomIndcs(a=10, b=10)
+        idx = g(8, 128)
+        assert idx.shape == (8, 10)
+        # no duplicates within a row
+        for r in idx:
+            assert len(set(r.tolist())) == 10
+        # rows differ (overwhelmingly likely)
+        assert not torch.equal(idx[0], idx[1]) or not torch.equal(idx[2], idx[3])
+
+    def test_batch_share(self):
+        torch.manual_seed(0)
+        g = GetRandomIndcs(a=5, b=5, is_batch_share=True)
+        idx = g(6, 64)
+        for r in range(1, 6):
+            assert torch.equal(idx[0], idx[r])
+
+    def test_beta_binomial_mean(self):
+        torch.manual_seed(0)
+        g = GetRandomIndcs(a=2.0, b=3.0, is_beta_binomial=True)
+        counts = [g(1, 100).shape[1] for _ in range(500)]
+        # E[BetaBinomial(100, 2, 3)] = 100 * 2/5 = 40
+        m = sum(counts) / len(counts)
+        assert 33 < m < 47, m
+
+    def test_stratified_counts_cycle(self):
+        from npf.utils.datasplit import StratifiedCountIndcs
+
+        s = StratifiedCountIndcs(a=0, b=4)
+        counts = [s(2, 16).shape[1] for _ in range(10)]
+        assert counts == [0, 1, 2, 3, 4, 0, 1, 2, 3, 4]
+        s.reset()
+        assert s(2, 16).shape[1] == 0
+
+    def test_range_indcs_window(self):
+        torch.manual_seed(0)
+        g = GetRandomIndcs(a=3, b=3, range_indcs=(10, 20))
+        idx = g(4, 128)
+        assert idx.min() >= 10 and idx.max() < 20
+
+    def test_grid_select_matches_manual(self):
+        from npf.utils.datasplit import GridCntxtTrgtGetter, RandomMasker, no_masker
+
+        torch.manual_seed(0)
+        getter = GridCntxtTrgtGetter(
+            context_masker=RandomMasker(a=5, b=5), target_masker=no_masker
+        )
+        X = torch.randn(2, 3, 8, 8)  # [B, C, H, W]
+        Xc, Yc, Xt, Yt = getter(X)
+        assert Xc.shape == (2, 5, 2) and Yc.shape == (2, 5, 3)
+        assert Xt.shape == (2, 64, 2) and Yt.shape == (2, 64, 3)
+        # target coords are the full normalized grid in row-major order
+        lin = torch.linspace(-1, 1, 8)
+        assert torch.allclose(Xt[0, :8, 1], lin)  # first row sweeps W
+        assert torch.allclose(Xt[0, ::8, 0], lin)  # first col sweeps H
+        # values match the image at those positions
+        Xl = X.movedim(1, -1).reshape(2, 64, 3)
+        assert torch.equal(Yt, Xl)
+
+    def test_upscale_factor(self):
+        from npf.utils.datasplit import GridCntxtTrgtGetter
+
+        getter = GridCntxtTrgtGetter(upscale_factor=2.0)
+        X = torch.randn(1, 1, 4, 4)
+        _, _, Xt, _ = getter(X)
+        assert float(Xt.max()) == 2.0 and float(Xt.min()) == -2.0
+
+
+@pytest.mark.gpu
+class TestSplitterOnDevice:
+    def test_episode_stays_on_gpu(self):
+        from npf.utils.datasplit import (
+            CntxtTrgtGetter,
+            GetRandomIndcs,
+            GridCntxtTrgtGetter,
+            get_all_indcs,
+        )
+
+        X = torch.randn(4, 64, 1, device="cuda")
+        y = torch.randn(4, 64, 1, device="cuda")
+        getter = CntxtTrgtGetter(
+            contexts_getter=GetRandomIndcs(a=0.1, b=0.5), targets_getter=get_all_indcs
+        )
+        out = getter(X, y)
+        assert all(t.is_cuda for t in out)
+
+        Xg = torch.randn(2, 3, 16, 16, device="cuda")
+        gout = GridCntxtTrgtGetter()(Xg)
+        assert all(t.is_cuda for t in gout)
