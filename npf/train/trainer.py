@@ -56,6 +56,26 @@ def predefined_split(valid_dataset):
     return split
 
 
+def _validate_episode(inputs):
+    """Training features must be rescaled to [-1,1] (reference
+    base.py:241-247).  The model skips the check on GPU tensors (a `.all()`
+    there forces a device sync every step and is illegal under hipGraph
+    capture), so the trainer enforces it HERE, on the loader's CPU tensors,
+    before the H2D copy — bad data fails loudly without touching the hot
+    path.  Device-resident episodes (GPU-side sampler) are trusted: the
+    sampler constructs coordinates in-range by design."""
+    for key in ("X_cntxt", "X_trgt"):
+        X = inputs.get(key) if isinstance(inputs, dict) else None
+        if X is not None and torch.is_tensor(X) and not X.is_cuda \
+                and torch.is_floating_point(X) and X.numel():
+            lo, hi = float(X.min()), float(X.max())
+            if lo < -1 - 1e-6 or hi > 1 + 1e-6:
+                raise ValueError(
+                    f"Features during training should be in [-1,1]: "
+                    f"{lo} <= {key} <= {hi}."
+                )
+
+
 def _move(obj, device, non_blocking=True):
     if torch.is_tensor(obj):
         return obj.to(device, non_blocking=non_blocking)
@@ -272,6 +292,7 @@ class NPFTrainer:
             if hasattr(loader.sampler, "set_epoch"):
                 loader.sampler.set_epoch(epoch)
             for inputs, y in loader:
+                _validate_episode(inputs)
                 inputs = _move(inputs, self.device)
                 y = _move(y, self.device)
                 loss = self.train_step(inputs, y)
@@ -377,17 +398,19 @@ def eval_loglike(trainer, dataset, seed=123):
     ll = torch.cat(all_ll, dim=0)
 
     if world > 1:
-        # DistributedSampler shards rank-interleaved: invert the permutation
+        # DistributedSampler (shuffle=False) pads the index list to a
+        # multiple of `world` by repeating its head, then deals it
+        # round-robin; invert exactly that assignment so rows come back in
+        # dataset order (padded duplicates collapse onto their slot).
         ll_dev = ll.to(trainer.device)
         gathered = dist_utils.all_gather_cat(ll_dev, dim=0).cpu()
         n_total = len(dataset)
-        order = []
+        shard_len = gathered.shape[0] // world
+        padded = np.arange(world * shard_len) % n_total
+        ll = torch.empty(n_total, dtype=gathered.dtype)
         for r in range(world):
-            order.extend(range(r, n_total, world))
-        # gathered rows are [rank0 shard; rank1 shard; ...]
-        inv = np.empty(len(order), dtype=np.int64)
-        inv[np.asarray(order[: gathered.shape[0]])] = np.arange(gathered.shape[0])
-        ll = gathered[inv[:n_total]]
+            for j in range(shard_len):
+                ll[padded[r + j * world]] = gathered[r * shard_len + j]
 
     return ll.numpy()
 

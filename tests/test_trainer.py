@@ -187,3 +187,32 @@ def test_dataset_merger_attr_forwarding():
     assert m.min_max == a.min_max  # attr forwarded from the first dataset
     x, y = m[5]
     assert x.shape == (8, 1)
+
+
+def test_trainer_validates_input_range_on_cpu():
+    """Out-of-range training features fail loudly at the loader boundary
+    (the model-side check is skipped on GPU for hipGraph safety)."""
+    from npf import CNPFLoss
+    from npf.train.trainer import NPFTrainer
+
+    class BadDS(torch.utils.data.Dataset):
+        def __len__(self):
+            return 8
+
+        def __getitem__(self, i):
+            return torch.randn(16, 1) * 3, torch.randn(16, 1)
+
+    def collate(batch):
+        X = torch.stack([b[0] for b in batch])
+        y = torch.stack([b[1] for b in batch])
+        inputs = dict(X_cntxt=X[:, :4], Y_cntxt=y[:, :4], X_trgt=X, Y_trgt=y)
+        return inputs, y
+
+    from npf.zoo import cnp_1d
+
+    tr = NPFTrainer(
+        cnp_1d(), CNPFLoss(), collate_fn=collate, device="cpu",
+        batch_size=4, max_epochs=1,
+    )
+    with pytest.raises(ValueError, match=r"\[-1,1\]"):
+        tr.fit(BadDS())
