@@ -174,10 +174,8 @@ def _configs():
             batch=32,
             pool=lambda dev, b, s: _img_point_pool(dev, b, s, shape=(3, 32, 32)),
             needs_y_trgt=True,  # NPVI: q_zCct from the target set
-            graph_ok=False,  # latent rsample inside a hipGraph replays
-                             # corrupted noise on ROCm -> loss NaN; the step
-                             # is compute-bound (~11 ms) so replay gains
-                             # nothing anyway
+            is_latent=True,  # rsample routed through the static noise pool
+                             # (npf.ops.noise) so the step captures cleanly
             desc="AttnLNP-2D CelebA32-shape (self-attn encoder, NPVI, 468,486 params)",
             seq_len=32 * 32,
             n_cntxt=int(0.3 * 32 * 32),
@@ -188,7 +186,7 @@ def _configs():
             batch=16,
             pool=lambda dev, b, s: _img_grid_pool(dev, b, s, shape=(3, 64, 64)),
             needs_y_trgt=False,  # NPML
-            graph_ok=False,  # see attnlnp2d
+            is_latent=True,  # see attnlnp2d
             desc="GridConvLNP-2D CelebA64-shape (4+4-block CNN k=9, NPML z=16, 487,793 params)",
             seq_len=64 * 64,
             n_cntxt=int(0.1 * 64 * 64),
@@ -307,9 +305,20 @@ def main():
         loss.backward()
         return loss
 
+    # latent models: draw z from the static noise pool so the graph can
+    # capture the whole step; the pool is refreshed outside the graph in
+    # load() every iteration (fresh noise per step, same training stats)
+    use_noise_pool = use_cuda and cfg.get("is_latent", False) and not args.no_graph
+    if use_noise_pool:
+        from npf.ops.noise import enable_noise_pool, refresh_noise_
+
+        enable_noise_pool(True)
+
     def load(i):
         for buf, t in zip((sXc, sYc, sXt, sYt), pool[i % POOL_BATCHES]):
             buf.copy_(t)
+        if use_noise_pool:
+            refresh_noise_()
 
     # ---- warmup (also primes BLAS/MIOpen algo caches) ----
     for i in range(max(args.warmup, 3)):

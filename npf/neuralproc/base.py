@@ -283,8 +283,27 @@ class LatentNeuralProcessFamily(NeuralProcessFamily):
             sampling_dist = q_zCc
 
         # [n_z_samples, B, *n_lat, z_dim]
-        z_samples = sampling_dist.rsample([self.n_z_samples])
+        z_samples = self._rsample(sampling_dist, self.n_z_samples)
         return z_samples, q_zCc, q_zCct
+
+    @staticmethod
+    def _rsample(dist, n_z_samples):
+        """rsample that stays valid inside a hipGraph capture.
+
+        With the noise pool enabled (npf.ops.noise), the reparameterized
+        draw is `loc + scale * eps` over a static pool buffer refreshed by
+        the host between replays — torch RNG kernels inside a capture would
+        replay frozen Philox state.  Default path: plain rsample.
+        """
+        from npf.ops import noise as _noise
+
+        base = getattr(dist, "base_dist", None)
+        if _noise.is_noise_pool_enabled() and base is not None:
+            eps = _noise.pool_noise(
+                (n_z_samples, *base.loc.shape), base.loc.device, base.loc.dtype
+            )
+            return base.loc + base.scale * eps
+        return dist.rsample([n_z_samples])
 
     def infer_latent_dist(self, X, R):
         """R -> q(z) (reference base.py:516-547)."""
