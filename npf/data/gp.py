@@ -167,12 +167,13 @@ class GPDataset(Dataset):
         )
         return data, targets
 
-    def sample_tasks(self, n_samples, n_points, min_max):
+    def sample_tasks(self, n_samples, n_points, min_max, out_device="cpu"):
         """Batched GP prior draws.
 
         Groups of `n_same_samples` functions share X and kernel
         hyperparameters (reference gaussian_process.py:202-231); all groups
-        are factorized with one batched Cholesky.
+        are factorized with one batched Cholesky.  `out_device` lets the
+        device-resident loader keep the epoch in HBM (no host round-trip).
         """
         device = self.device
         n_groups = (n_samples + self.n_same_samples - 1) // self.n_same_samples
@@ -200,11 +201,11 @@ class GPDataset(Dataset):
         )
         Y = Y.permute(0, 2, 1).unsqueeze(-1)  # [G, S, N, 1]
 
-        X = X.reshape(-1, n_points, 1)[:n_samples].cpu()
-        Y = Y.reshape(-1, n_points, 1)[:n_samples].contiguous().cpu()
+        X = X.reshape(-1, n_points, 1)[:n_samples].to(out_device)
+        Y = Y.reshape(-1, n_points, 1)[:n_samples].contiguous().to(out_device)
 
         # shuffle so same-group functions are not consecutive
-        perm = torch.randperm(n_samples)
+        perm = torch.randperm(n_samples, device=X.device)
         X, Y = X[perm], Y[perm]
 
         X = rescale_range(X, self.min_max, (-1, 1)).float()

@@ -212,6 +212,17 @@ class NPFTrainer:
     def _loader(self, dataset, training):
         from torch.utils.data.distributed import DistributedSampler
 
+        from npf.train.device_loader import DeviceEpisodes
+
+        if isinstance(dataset, DeviceEpisodes):
+            # device-resident episodes: no DataLoader, no collate, no H2D;
+            # fresh-task ranks draw from their own RNG stream (rank-offset
+            # seed), so no sampler sharding either
+            return dataset.batches(
+                self.batch_size if training else self.valid_batch_size,
+                training=training,
+            )
+
         sampler = None
         shuffle = self.shuffle if training else False
         if self.world_size > 1 and getattr(dataset, "is_rank_sharded", False) is False:
@@ -289,7 +300,7 @@ class NPFTrainer:
             n_tasks = 0
             train_losses = []
             loader = self._loader(dataset, training=True)
-            if hasattr(loader.sampler, "set_epoch"):
+            if hasattr(getattr(loader, "sampler", None), "set_epoch"):
                 loader.sampler.set_epoch(epoch)
             for inputs, y in loader:
                 _validate_episode(inputs)
@@ -396,6 +407,13 @@ def eval_loglike(trainer, dataset, seed=123):
         all_ll.append(-loss.float().cpu())
     trainer.criterion.reduction = old_reduction
     ll = torch.cat(all_ll, dim=0)
+
+    from npf.train.device_loader import DeviceEpisodes
+
+    if isinstance(dataset, DeviceEpisodes):
+        # device episodes are not rank-sharded: every rank evaluated the
+        # whole set already
+        return ll.numpy()
 
     if world > 1:
         # DistributedSampler (shuffle=False) pads the index list to a

@@ -216,3 +216,88 @@ def test_trainer_validates_input_range_on_cpu():
     )
     with pytest.raises(ValueError, match=r"\[-1,1\]"):
         tr.fit(BadDS())
+
+
+class TestDeviceEpisodes:
+    """Device-resident episode loader (npf/train/device_loader.py)."""
+
+    def _ds(self, n=32, reuse=True):
+        from npf.data import GPDataset
+        from npf.data.kernels import RBF
+
+        set_seed(5)
+        return GPDataset(
+            kernel=RBF(0.2), n_samples=n, n_points=16,
+            is_reuse_across_epochs=reuse,
+        )
+
+    def test_fit_and_eval_run(self):
+        from npf import CNPFLoss
+        from npf.train.device_loader import DeviceEpisodes
+        from npf.train.trainer import NPFTrainer, eval_loglike
+        from npf.utils.datasplit import CntxtTrgtGetter, GetRandomIndcs
+        from npf.zoo import cnp_1d
+
+        ds = self._ds(reuse=False)
+        eps = DeviceEpisodes(
+            ds, CntxtTrgtGetter(contexts_getter=GetRandomIndcs(a=2, b=6)),
+            device="cpu",
+        )
+        tr = NPFTrainer(
+            cnp_1d(), CNPFLoss(), device="cpu", batch_size=8, max_epochs=2,
+            seed=0,
+        )
+        tr.fit(eps)
+        assert len(tr.history) == 2
+        assert tr.history[0]["train_loss"] is not None
+        ll = eval_loglike(tr, DeviceEpisodes(
+            self._ds(), CntxtTrgtGetter(contexts_getter=GetRandomIndcs(a=2, b=6)),
+            device="cpu"), seed=123)
+        assert ll.shape == (32,)
+
+    def test_eval_matches_dataloader_path(self):
+        """Same frozen tasks + deterministic splitter: the device-resident
+        eval equals the DataLoader/collate eval per task."""
+        import numpy as np
+
+        from npf import CNPFLoss
+        from npf.data import cntxt_trgt_collate
+        from npf.train.device_loader import DeviceEpisodes
+        from npf.train.trainer import NPFTrainer, eval_loglike
+        from npf.utils.datasplit import (
+            CntxtTrgtGetter,
+            GetRangeIndcs,
+            get_all_indcs,
+        )
+        from npf.zoo import cnp_1d
+
+        ds = self._ds()
+        splitter = CntxtTrgtGetter(
+            contexts_getter=GetRangeIndcs((0, 5)), targets_getter=get_all_indcs
+        )
+        model = cnp_1d()
+
+        tr1 = NPFTrainer(
+            model, CNPFLoss(), device="cpu", batch_size=8,
+            collate_fn=cntxt_trgt_collate(splitter),
+        )
+        ll_loader = eval_loglike(tr1, ds, seed=123)
+
+        tr2 = NPFTrainer(model, CNPFLoss(), device="cpu", batch_size=8)
+        ll_device = eval_loglike(
+            tr2, DeviceEpisodes(ds, splitter, device="cpu"), seed=123
+        )
+        assert np.allclose(ll_loader, ll_device, atol=1e-5)
+
+    def test_fresh_epochs_regenerate(self):
+        from npf.train.device_loader import DeviceEpisodes
+        from npf.utils.datasplit import CntxtTrgtGetter, GetRangeIndcs
+
+        ds = self._ds(reuse=False)
+        eps = DeviceEpisodes(
+            ds, CntxtTrgtGetter(contexts_getter=GetRangeIndcs((0, 4))),
+            device="cpu",
+        )
+        first = [y.clone() for _, y in eps.batches(8, training=True)]
+        second = [y.clone() for _, y in eps.batches(8, training=True)]
+        assert not torch.equal(first[0], second[0])
