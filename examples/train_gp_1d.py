@@ -44,18 +44,20 @@ def main():
     p.add_argument("--n-tasks", type=int, default=50000)
     p.add_argument("--chckpnt-dir", default="results/pretrained/")
     p.add_argument("--bf16", action="store_true")
+    p.add_argument("--device-episodes", action="store_true",
+                   help="GPU-resident tasks + on-device splitting (fast path)")
+    p.add_argument("--runs-suffix", default="")
     args = p.parse_args()
 
     train, test, valid = get_datasets_single_gp(n_samples=args.n_tasks)
     train = {k: v for k, v in train.items() if k in args.datasets}
 
     # reference 1D splitter: U(0, 50) contexts, all 128 points as targets
-    collate = cntxt_trgt_collate(
-        CntxtTrgtGetter(
-            contexts_getter=GetRandomIndcs(a=0.0, b=50),
-            targets_getter=get_all_indcs,
-        )
+    splitter = CntxtTrgtGetter(
+        contexts_getter=GetRandomIndcs(a=0.0, b=50),
+        targets_getter=get_all_indcs,
     )
+    collate = cntxt_trgt_collate(splitter)
 
     for name in args.models:
         builder, criterion, extra = MODELS[name]
@@ -76,6 +78,8 @@ def main():
             iterator_valid__collate_fn=collate,
             amp_dtype="bfloat16" if args.bf16 else None,
             grad_clip_norm=extra.get("grad_clip_norm"),
+            device_episodes=splitter if args.device_episodes else None,
+            is_progressbar=True,
         )
 
 

@@ -456,6 +456,7 @@ def train_models(
     seed=None,
     datasets_kwargs=dict(),
     models_kwargs=dict(),
+    device_episodes=None,
     **kwargs,
 ):
     """Grid-train {datasets} x {models} x runs (reference train.py:34-305).
@@ -465,6 +466,12 @@ def train_models(
     writes the same per-run directory scheme:
     `{chckpnt_dirname}{data}/{model}/run_{k}/{params.pt,optimizer.pt,
     model_summary.txt,eval.csv}`.
+
+    `device_episodes`: a CntxtTrgtGetter — wraps each training dataset in
+    DeviceEpisodes (tasks resident on the GPU, on-device splitting, fresh
+    epochs regenerated by the batched GPU sampler) instead of the
+    DataLoader/collate path.  Evaluation keeps the collate path (protocol
+    fidelity; it is off the hot path).
     """
     trainers = dict()
 
@@ -528,7 +535,14 @@ def train_models(
                     trainer.load_params()
 
                 if is_retrain:
-                    trainer.fit(data_train)
+                    fit_data = data_train
+                    if device_episodes is not None:
+                        from npf.train.device_loader import DeviceEpisodes
+
+                        fit_data = DeviceEpisodes(
+                            data_train, device_episodes, device=trainer.device
+                        )
+                    trainer.fit(fit_data)
                     if run_dir and dist_utils.get_rank() == 0:
                         with open(os.path.join(run_dir, MOD_SUMM_FILENAME), "w") as f:
                             f.write(str(trainer.module))
