@@ -213,6 +213,11 @@ def main():
 
     rank, world, local_rank = dist_utils.init_distributed()
     use_cuda = torch.cuda.is_available()
+    if use_cuda:
+        # modulo so an N-rank RCCL smoke run fits on fewer GPUs (e.g. two
+        # ranks sharing cuda:0 on a 1-GPU box to exercise the collectives)
+        local_rank = local_rank % torch.cuda.device_count()
+        torch.cuda.set_device(local_rank)
     device = torch.device(f"cuda:{local_rank}" if use_cuda else "cpu")
     if world > 1:
         assert world == args.gpus or args.gpus == 1, (world, args.gpus)
