@@ -49,7 +49,9 @@ def main():
     p.add_argument("--runs-suffix", default="")
     args = p.parse_args()
 
-    train, test, valid = get_datasets_single_gp(n_samples=args.n_tasks)
+    train, test, valid = get_datasets_single_gp(
+        n_samples=args.n_tasks, defer_generation=args.device_episodes
+    )
     train = {k: v for k, v in train.items() if k in args.datasets}
 
     # reference 1D splitter: U(0, 50) contexts, all 128 points as targets

@@ -88,6 +88,7 @@ class GPDataset(Dataset):
         is_reuse_across_epochs=True,
         device="cpu",
         generator=None,
+        defer_generation=False,
         **kwargs,
     ):
         self.n_samples = n_samples
@@ -107,7 +108,11 @@ class GPDataset(Dataset):
         self._idx_chunk = 0
         self._generator = generator
 
-        self.precompute_chunk_()
+        self.data = self.targets = None
+        if not defer_generation:
+            # eager generation serves the DataLoader path; the
+            # device-resident loader regenerates on-GPU and never reads it
+            self.precompute_chunk_()
 
     @property
     def generator(self):
@@ -132,6 +137,8 @@ class GPDataset(Dataset):
         return self.n_samples
 
     def __getitem__(self, index):
+        if self.data is None:
+            self.precompute_chunk_()
         if self.is_reuse_across_epochs:
             return self.data[index], self.targets[index]
         # fresh functions: serve sequentially, regenerate when exhausted
@@ -156,7 +163,14 @@ class GPDataset(Dataset):
 
         try:
             loaded = load_chunk({"data", "targets"}, save_file, idx_chunk)
-            return loaded["data"], loaded["targets"]
+            if loaded["data"].shape[0] == n_samples and loaded["data"].shape[1] == n_points:
+                return loaded["data"], loaded["targets"]
+            # stale cache from a different-sized run: regenerate (a smaller
+            # chunk silently served as a full epoch corrupts the budget)
+            self.logger.warning(
+                "cached chunk %s has %s tasks, need %s: regenerating",
+                idx_chunk, tuple(loaded["data"].shape), (n_samples, n_points),
+            )
         except NotLoadedError:
             pass
 
