@@ -52,6 +52,10 @@ void npf_gauss_ll_fwd_launch(const float*, const float*, const float*, float*,
 void npf_gauss_ll_bwd_launch(const float*, const float*, const float*,
                              const float*, float*, float*, long long,
                              long long, hipStream_t);
+void npf_lse_z_fwd_launch(const float*, float*, long long, long long,
+                          hipStream_t);
+void npf_lse_z_bwd_launch(const float*, const float*, const float*, float*,
+                          long long, long long, hipStream_t);
 void npf_cb_stats_launch(const void*, int, float*, float*, float*, float*,
                          float*, float*, int, int, int, float, float,
                          hipStream_t);
@@ -260,6 +264,33 @@ std::tuple<torch::Tensor, torch::Tensor> gauss_ll_bwd(torch::Tensor loc,
                           dloc.data_ptr<float>(), dscale.data_ptr<float>(),
                           rows, m, stream);
   return {dloc, dscale};
+}
+
+// NPML epilogue: logsumexp over the z dim of a [Z, B] matrix, minus log Z
+torch::Tensor lse_z_fwd(torch::Tensor w) {
+  check_cuda_contig(w, "w");
+  TORCH_CHECK(w.dim() == 2 && w.scalar_type() == torch::kFloat32,
+              "lse_z expects fp32 [Z, B]");
+  const long long Z = w.size(0), B = w.size(1);
+  auto out = torch::empty({B}, w.options());
+  auto stream = at::hip::getCurrentHIPStream();
+  npf_lse_z_fwd_launch(w.data_ptr<float>(), out.data_ptr<float>(), Z, B,
+                       stream);
+  return out;
+}
+
+torch::Tensor lse_z_bwd(torch::Tensor w, torch::Tensor out,
+                        torch::Tensor dout) {
+  check_cuda_contig(w, "w");
+  check_cuda_contig(out, "out");
+  check_cuda_contig(dout, "dout");
+  const long long Z = w.size(0), B = w.size(1);
+  auto dw = torch::empty_like(w);
+  auto stream = at::hip::getCurrentHIPStream();
+  npf_lse_z_bwd_launch(w.data_ptr<float>(), out.data_ptr<float>(),
+                       dout.data_ptr<float>(), dw.data_ptr<float>(), Z, B,
+                       stream);
+  return dw;
 }
 
 // fused conv block: stats (training BN) -> (mean, rstd, save_var)
@@ -579,6 +610,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("setconv_bwd", &setconv_bwd, "fused Gaussian SetConv backward");
   m.def("gauss_ll_fwd", &gauss_ll_fwd, "fused Gaussian log-lik forward");
   m.def("gauss_ll_bwd", &gauss_ll_bwd, "fused Gaussian log-lik backward");
+  m.def("lse_z_fwd", &lse_z_fwd, "NPML logmeanexp-over-z forward");
+  m.def("lse_z_bwd", &lse_z_bwd, "NPML logmeanexp-over-z backward");
   m.def("convblock_stats", &convblock_stats,
         "fused conv block: per-channel batch stats (+running update)");
   m.def("convblock_fwd", &convblock_fwd,

@@ -52,6 +52,58 @@ npf_gauss_ll_bwd(const float* __restrict__ loc, const float* __restrict__ scale,
   }
 }
 
+// ---------------------------------------------------------------------------
+// NPML epilogue: out[b] = logsumexp_z w[z,b] - log Z  (SURVEY.md §2.3 "NPML
+// objective" row; reference npf/losses.py:169-203 does this with ~6 torch
+// kernels).  Z <= 32 in every shipped config: one wave per b, each lane owns
+// a z stripe, shuffle-reduce max then sum.
+// ---------------------------------------------------------------------------
+
+extern "C" __global__ void __launch_bounds__(64)
+npf_lse_z_fwd(const float* __restrict__ w, float* __restrict__ out,
+              long long Z, long long B) {
+  const long long b = blockIdx.x;
+  if (b >= B) return;
+  float mx = -INFINITY;
+  for (long long z = threadIdx.x; z < Z; z += 64)
+    mx = fmaxf(mx, w[z * B + b]);
+  for (int off = 32; off; off >>= 1)
+    mx = fmaxf(mx, __shfl_down(mx, off));
+  mx = __shfl(mx, 0);
+  float s = 0.f;
+  for (long long z = threadIdx.x; z < Z; z += 64)
+    s += __expf(w[z * B + b] - mx);
+  for (int off = 32; off; off >>= 1)
+    s += __shfl_down(s, off);
+  if (threadIdx.x == 0) out[b] = mx + __logf(s) - __logf((float)Z);
+}
+
+// dw[z,b] = dout[b] * softmax_z(w)[z,b] = dout[b] * exp(w - (out[b] + log Z))
+extern "C" __global__ void __launch_bounds__(256)
+npf_lse_z_bwd(const float* __restrict__ w, const float* __restrict__ out,
+              const float* __restrict__ dout, float* __restrict__ dw,
+              long long Z, long long B) {
+  const long long i = (long long)blockIdx.x * 256 + threadIdx.x;
+  if (i >= Z * B) return;
+  const long long b = i % B;
+  const float lse = out[b] + __logf((float)Z);
+  dw[i] = dout[b] * __expf(w[i] - lse);
+}
+
+extern "C" void npf_lse_z_fwd_launch(const float* w, float* out, long long Z,
+                                     long long B, hipStream_t stream) {
+  hipLaunchKernelGGL(npf_lse_z_fwd, dim3((unsigned)B), dim3(64), 0, stream, w,
+                     out, Z, B);
+}
+
+extern "C" void npf_lse_z_bwd_launch(const float* w, const float* out,
+                                     const float* dout, float* dw, long long Z,
+                                     long long B, hipStream_t stream) {
+  const long long n = Z * B;
+  hipLaunchKernelGGL(npf_lse_z_bwd, dim3((unsigned)((n + 255) / 256)),
+                     dim3(256), 0, stream, w, out, dout, dw, Z, B);
+}
+
 extern "C" void npf_gauss_ll_fwd_launch(const float* loc, const float* scale,
                                         const float* y, float* out,
                                         long long rows, long long m,

@@ -127,6 +127,19 @@ class NLLLossLNPF(BaseLossNPF):
     def get_loss(self, p_yCc, z_samples, q_zCc, q_zCct, Y_trgt):
         n_z_samples = p_yCc.batch_shape[0]
 
+        if (
+            q_zCct is None
+            and isinstance(p_yCc, Independent)
+            and isinstance(p_yCc.base_dist, Normal)
+            and p_yCc.reinterpreted_batch_ndims == 1
+            and Y_trgt.dim() == p_yCc.base_dist.loc.dim() - 1
+        ):
+            # no importance weights: the whole objective (per-z target-summed
+            # log-lik + logmeanexp over z) is one fused kernel pair
+            return -ops.gaussian_nll_logmeanexp(
+                p_yCc.base_dist.loc, p_yCc.base_dist.scale, Y_trgt
+            )
+
         # [Z, B]
         sum_log_w_k = sum_log_prob(p_yCc, Y_trgt)
         if q_zCct is not None:

@@ -12,7 +12,7 @@ import torch
 
 from . import _backend
 
-__all__ = ["attention_qkv", "setconv_gaussian", "gaussian_nll_sum", "conv_block_1d", "conv_block_2d", "grid_density", "gaussian_kl_sum", "mlp_chain"]
+__all__ = ["attention_qkv", "setconv_gaussian", "gaussian_nll_sum", "gaussian_nll_logmeanexp", "conv_block_1d", "conv_block_2d", "grid_density", "gaussian_kl_sum", "mlp_chain"]
 
 
 # --------------------------------------------------------------------------- #
@@ -253,6 +253,49 @@ def gaussian_nll_sum(loc, scale, y):
     if loc.is_cuda and loc.dtype == torch.float32:
         return _GaussLLFn.apply(loc.contiguous(), scale.contiguous(), y.contiguous())
     return _nll_ref(loc, scale, y)
+
+
+class _GaussLLLseFn(torch.autograd.Function):
+    """NPML objective fused end-to-end: per-z target-summed log-lik then
+    logmeanexp over z, [Z,B,...,Y] -> [B] (SURVEY.md §2.3 "NPML objective";
+    reference losses.py:169-203 runs ~6 extra kernels for the logsumexp)."""
+
+    @staticmethod
+    def forward(ctx, loc, scale, y):
+        ext = _backend.require_extension("gaussian_nll_logmeanexp")
+        w = ext.gauss_ll_fwd(loc, scale, y)  # [Z, B]
+        out = ext.lse_z_fwd(w)  # [B]
+        ctx.save_for_backward(loc, scale, y, w, out)
+        return out
+
+    @staticmethod
+    def backward(ctx, dout):
+        ext = _backend.extension()
+        loc, scale, y, w, out = ctx.saved_tensors
+        dw = ext.lse_z_bwd(w, out, dout.contiguous())
+        dloc, dscale = ext.gauss_ll_bwd(loc, scale, y, dw)
+        return dloc, dscale, None
+
+
+def gaussian_nll_logmeanexp(loc, scale, y):
+    """log mean_z exp( sum_targets log N(y; loc, scale) ): [Z,B,..,Y] -> [B].
+
+    The NPML training/eval objective without importance weights
+    (q_zCct=None), one fused kernel pair instead of NLL + torch logsumexp.
+    """
+    if y.dim() == loc.dim() - 1:
+        y = y.unsqueeze(0)
+    y = y.expand_as(loc)
+    if loc.dtype not in (torch.float32, torch.float64):
+        loc, scale = loc.float(), scale.float()
+    y = y.to(loc.dtype)
+    if loc.is_cuda and loc.dtype == torch.float32 and loc.dim() >= 2 \
+            and _backend.require_extension("gaussian_nll_logmeanexp") is not None:
+        return _GaussLLLseFn.apply(
+            loc.contiguous(), scale.contiguous(), y.contiguous()
+        )
+    w = _nll_ref(loc, scale, y)
+    return torch.logsumexp(w, 0) - math.log(w.shape[0])
 
 
 # --------------------------------------------------------------------------- #

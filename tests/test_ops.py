@@ -612,3 +612,78 @@ class TestZsmmsFusedParity:
         assert torch.allclose(out.cpu(), out0, atol=5e-4), (
             (out.cpu() - out0).abs().max()
         )
+
+
+class TestNLLLogMeanExp:
+    """Fused NPML objective (gaussian_nll_logmeanexp) vs composed torch."""
+
+    def test_cpu_reference_matches_composed(self):
+        import math
+
+        from npf.ops import gaussian_nll_logmeanexp
+
+        torch.manual_seed(0)
+        Z, B, T, Y = 8, 6, 17, 2
+        loc = torch.randn(Z, B, T, Y, requires_grad=True)
+        scale = torch.rand(Z, B, T, Y).add(0.3).requires_grad_()
+        y = torch.randn(B, T, Y)
+        out = gaussian_nll_logmeanexp(loc, scale, y)
+
+        import torch.distributions as D
+
+        p = D.Independent(D.Normal(loc, scale), 1)
+        w = p.log_prob(y.unsqueeze(0).expand(Z, B, T, Y)).sum(-1)
+        ref = torch.logsumexp(w, 0) - math.log(Z)
+        assert torch.allclose(out, ref, atol=1e-5)
+        out.sum().backward()
+        g1, g2 = loc.grad.clone(), scale.grad.clone()
+        loc.grad = scale.grad = None
+        ref.sum().backward()
+        assert torch.allclose(g1, loc.grad, atol=1e-5)
+        assert torch.allclose(g2, scale.grad, atol=1e-5)
+
+    def test_nll_loss_dispatch_equals_manual(self):
+        import math
+
+        import torch.distributions as D
+
+        from npf import NLLLossLNPF
+
+        torch.manual_seed(1)
+        Z, B, T, Y = 4, 5, 9, 1
+        loc = torch.randn(Z, B, T, Y)
+        scale = torch.rand(Z, B, T, Y).add(0.3)
+        yt = torch.randn(B, T, Y)
+        p = D.Independent(D.Normal(loc, scale), 1)
+        crit = NLLLossLNPF(reduction=None)
+        crit.eval()
+        loss = crit((p, None, None, None), yt)
+        w = p.log_prob(yt.unsqueeze(0).expand(Z, B, T, Y)).sum(-1)
+        ref = -(torch.logsumexp(w, 0) - math.log(Z))
+        assert torch.allclose(loss, ref, atol=1e-5)
+
+
+@pytest.mark.gpu
+class TestNLLLogMeanExpGPU:
+    def test_fused_matches_fp32_reference(self):
+        import math
+
+        from npf.ops import gaussian_nll_logmeanexp
+
+        torch.manual_seed(0)
+        Z, B, T, Y = 16, 32, 128, 1
+        loc = torch.randn(Z, B, T, Y, device="cuda", requires_grad=True)
+        scale = torch.rand(Z, B, T, Y, device="cuda").add(0.3).requires_grad_()
+        y = torch.randn(B, T, Y, device="cuda")
+        out = gaussian_nll_logmeanexp(loc, scale, y)
+        out.sum().backward()
+
+        loc0 = loc.detach().cpu().requires_grad_()
+        scale0 = scale.detach().cpu().requires_grad_()
+        out0 = gaussian_nll_logmeanexp(loc0, scale0, y.cpu())
+        out0.sum().backward()
+        assert torch.allclose(out.cpu(), out0, atol=1e-4), (
+            (out.cpu() - out0).abs().max()
+        )
+        assert torch.allclose(loc.grad.cpu(), loc0.grad, atol=1e-4)
+        assert torch.allclose(scale.grad.cpu(), scale0.grad, atol=1e-4)
