@@ -46,6 +46,8 @@ def main():
     p.add_argument("--bf16", action="store_true")
     p.add_argument("--device-episodes", action="store_true",
                    help="GPU-resident tasks + on-device splitting (fast path)")
+    p.add_argument("--hipgraphs", action="store_true",
+                   help="capture whole optimization steps per episode shape")
     p.add_argument("--runs-suffix", default="")
     args = p.parse_args()
 
@@ -81,6 +83,7 @@ def main():
             amp_dtype="bfloat16" if args.bf16 else None,
             grad_clip_norm=extra.get("grad_clip_norm"),
             device_episodes=splitter if args.device_episodes else None,
+            hipgraphs=args.hipgraphs,
             is_progressbar=True,
         )
 

@@ -46,8 +46,11 @@ def init_distributed(backend=None, timeout_s=600):
         dist.init_process_group(backend, timeout=timedelta(seconds=timeout_s))
     rank = dist.get_rank()
     world = dist.get_world_size()
-    local_rank = int(os.environ.get("LOCAL_RANK", rank % max(torch.cuda.device_count(), 1)))
+    local_rank = int(os.environ.get("LOCAL_RANK", rank))
     if torch.cuda.is_available():
+        # modulo: an N-rank smoke run may share fewer GPUs (e.g. 2 ranks on
+        # one box to exercise RCCL init/collectives)
+        local_rank = local_rank % torch.cuda.device_count()
         torch.cuda.set_device(local_rank)
     return rank, world, local_rank
 

@@ -135,6 +135,7 @@ class NPFTrainer:
         num_workers=0,
         is_progressbar=False,
         profile=False,
+        hipgraphs=False,
     ):
         self.rank = dist_utils.get_rank()
         self.world_size = dist_utils.get_world_size()
@@ -178,13 +179,33 @@ class NPFTrainer:
         # ranges delimit forward/backward/optimizer per-kernel attribution)
         self.profile = profile and torch.cuda.is_available()
 
-        self.optimizer = optimizer(self.module.parameters(), lr=lr)
+        # MI355X hot path: fused multi-tensor Adam (one kernel instead of
+        # ~10 foreach launches), capturable with a device lr tensor so the
+        # LR schedule survives hipGraph replay
+        self.hipgraphs = bool(hipgraphs) and self.device.type == "cuda"
+        self._lr_tensor = None
+        if self.device.type == "cuda" and optimizer is Adam:
+            self._lr_tensor = torch.tensor(float(lr), device=self.device)
+            try:
+                self.optimizer = Adam(
+                    self.module.parameters(), lr=self._lr_tensor,
+                    fused=True, capturable=True,
+                )
+            except Exception:
+                self._lr_tensor = None
+                self.optimizer = optimizer(self.module.parameters(), lr=lr)
+        else:
+            self.optimizer = optimizer(self.module.parameters(), lr=lr)
         self.scheduler = None
+        self._lr_gamma = None
         if decay_lr is not None:
             gamma = get_exponential_decay_gamma(decay_lr, max_epochs)
-            self.scheduler = torch.optim.lr_scheduler.ExponentialLR(
-                self.optimizer, gamma=gamma
-            )
+            self._lr_gamma = gamma
+            if self._lr_tensor is None:
+                self.scheduler = torch.optim.lr_scheduler.ExponentialLR(
+                    self.optimizer, gamma=gamma
+                )
+        self._step_graphs = None
 
         self.ddp = None
         if is_ddp and dist_utils.is_distributed():
@@ -251,26 +272,16 @@ class NPFTrainer:
         if self.profile:
             torch.cuda.nvtx.range_pop()
 
-    def train_step(self, inputs, y):
-        """One optimization step; returns the (scalar) loss."""
-        self.module.train()
-        self.criterion.train()
+    def _step_body(self, inputs, y):
+        """fwd + loss + bwd + reduce + clip + optimizer — replay-safe."""
         if self.ddp is not None:
             self.ddp.zero_grad_()
         else:
             self.optimizer.zero_grad(set_to_none=True)
-
-        self._range_push("npf/forward")
         with self._autocast():
             outputs = self.module(**inputs)
         loss = self.criterion(outputs, y)  # loss math in fp32
-        self._range_pop()
-
-        self._range_push("npf/backward")
         loss.backward()
-        self._range_pop()
-
-        self._range_push("npf/optimizer")
         if self.ddp is not None:
             self.ddp.reduce_()
         if self.grad_clip_norm is not None:
@@ -278,8 +289,39 @@ class NPFTrainer:
                 self.module.parameters(), self.grad_clip_norm
             )
         self.optimizer.step()
-        self._range_pop()
         return loss.detach()
+
+    def _maybe_graph_step(self, inputs, y):
+        """Replay the step as one hipGraph when eligible (fixed-address
+        buffers, RNG via the noise pool); returns None to run eager."""
+        if not self.hipgraphs:
+            return None
+        if self._step_graphs is None:
+            # the first eager step (run by the caller) has primed optimizer
+            # state; SUMO-style stochastic z counts are not graph-safe
+            n_z = getattr(self.module, "n_z_samples_train", None)
+            if hasattr(n_z, "rvs"):
+                self.hipgraphs = False
+                return None
+            from npf.train.step_graphs import GraphedStepper
+
+            self._step_graphs = GraphedStepper(
+                self.module, self.optimizer, self._step_body
+            )
+        return self._step_graphs.step(inputs, y)
+
+    def train_step(self, inputs, y, _first=False):
+        """One optimization step; returns the loss (0-dim device tensor)."""
+        self.module.train()
+        self.criterion.train()
+        self._range_push("npf/step")
+        loss = None if _first else self._maybe_graph_step(inputs, y)
+        if loss is None:
+            loss = self._step_body(inputs, y)
+        else:
+            loss = loss.detach().clone()  # static buffer: next replay overwrites
+        self._range_pop()
+        return loss
 
     def validation_step(self, inputs, y):
         self.module.eval()
@@ -302,21 +344,30 @@ class NPFTrainer:
             loader = self._loader(dataset, training=True)
             if hasattr(getattr(loader, "sampler", None), "set_epoch"):
                 loader.sampler.set_epoch(epoch)
+            first = epoch == 0
             for inputs, y in loader:
                 _validate_episode(inputs)
                 inputs = _move(inputs, self.device)
                 y = _move(y, self.device)
-                loss = self.train_step(inputs, y)
-                train_losses.append(float(loss))
+                # losses stay on-device until epoch end: a float() here
+                # would sync the pipeline every step
+                loss = self.train_step(inputs, y, _first=first)
+                first = False
+                train_losses.append(loss)
                 n_tasks += y.shape[0] * self.world_size
+            if self.device.type == "cuda":
+                torch.cuda.synchronize()
             dur = time.perf_counter() - t0
 
             record = {
                 "epoch": epoch + 1,
-                "train_loss": float(np.mean(train_losses)) if train_losses else None,
+                "train_loss": (
+                    float(torch.stack(train_losses).float().mean())
+                    if train_losses else None
+                ),
                 "dur": dur,
                 "tasks_per_sec": n_tasks / dur if dur > 0 else None,
-                "lr": self.optimizer.param_groups[0]["lr"],
+                "lr": float(self.optimizer.param_groups[0]["lr"]),
             }
 
             if valid_dataset is not None:
@@ -346,6 +397,10 @@ class NPFTrainer:
 
             if self.scheduler is not None:
                 self.scheduler.step()
+            elif self._lr_gamma is not None and self._lr_tensor is not None:
+                # write the decayed lr INTO the device tensor the (possibly
+                # graph-captured) fused Adam reads on every step
+                self._lr_tensor.fill_(self.lr * self._lr_gamma ** (epoch + 1))
 
             if self.is_progressbar and self.rank == 0:
                 msg = f"epoch {epoch + 1}/{self.max_epochs} " + " ".join(
