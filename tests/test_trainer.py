@@ -301,3 +301,58 @@ class TestDeviceEpisodes:
         first = [y.clone() for _, y in eps.batches(8, training=True)]
         second = [y.clone() for _, y in eps.batches(8, training=True)]
         assert not torch.equal(first[0], second[0])
+
+
+@pytest.mark.gpu
+class TestGraphedStepEquivalence:
+    def _mk(self, hipgraphs):
+        from npf import CNPFLoss
+        from npf.train.trainer import NPFTrainer
+        from npf.zoo import attncnp_1d
+
+        torch.manual_seed(0)
+        tr = NPFTrainer(
+            attncnp_1d(), CNPFLoss(), device="cuda", batch_size=8,
+            amp_dtype=torch.bfloat16, hipgraphs=hipgraphs, seed=0,
+        )
+        return tr
+
+    def _episodes(self, n_steps, seed=7):
+        g = torch.Generator().manual_seed(seed)
+        eps = []
+        for i in range(n_steps):
+            n_c = 3 + (i % 3) * 7  # a few distinct shapes, each seen twice+
+            Xc = (torch.rand(8, n_c, 1, generator=g) * 2 - 1).cuda()
+            Yc = torch.randn(8, n_c, 1, generator=g).cuda()
+            Xt = (torch.rand(8, 64, 1, generator=g) * 2 - 1).cuda()
+            Yt = torch.randn(8, 64, 1, generator=g).cuda()
+            eps.append((dict(X_cntxt=Xc, Y_cntxt=Yc, X_trgt=Xt, Y_trgt=Yt), Yt))
+        return eps
+
+    def test_graphed_steps_match_eager_steps(self):
+        """The per-shape captured step must produce the same parameter
+        trajectory as the eager step (same episodes, same init)."""
+        eps = self._episodes(12)
+        t_e = self._mk(hipgraphs=False)
+        t_g = self._mk(hipgraphs=True)
+        # identical initial parameters
+        t_g.module.load_state_dict(t_e.module.state_dict())
+
+        losses_e, losses_g = [], []
+        for i, (inputs, y) in enumerate(eps):
+            losses_e.append(float(t_e.train_step(inputs, y, _first=i == 0)))
+        for i, (inputs, y) in enumerate(eps):
+            losses_g.append(float(t_g.train_step(inputs, y, _first=i == 0)))
+        torch.cuda.synchronize()
+
+        import numpy as np
+
+        assert np.allclose(losses_e, losses_g, rtol=1e-3, atol=1e-3), (
+            list(zip(losses_e, losses_g))
+        )
+        for (n1, p1), (n2, p2) in zip(
+            t_e.module.named_parameters(), t_g.module.named_parameters()
+        ):
+            assert torch.allclose(p1, p2, atol=1e-4), (
+                n1, (p1 - p2).abs().max()
+            )
