@@ -33,6 +33,7 @@ def build():
         os.path.join(REPO, "csrc", "npf_hip", "griddensity.hip"),
         os.path.join(REPO, "csrc", "npf_hip", "gauss_kl.hip"),
         os.path.join(REPO, "csrc", "npf_hip", "mlp_chain.hip"),
+        os.path.join(REPO, "csrc", "npf_hip", "attender.hip"),
     ]
     ext = CUDAExtension(
         name="npf._hip_C",

@@ -33,14 +33,15 @@ typedef __attribute__((ext_vector_type(8))) short qk_bf16x8;
 typedef __attribute__((ext_vector_type(4))) float qk_f32x4;
 
 struct QkvProblem {
-  const __hip_bfloat16* x;  // [R, D] row-major
-  const float* w;           // [D, D] torch Linear weight (out, in)
-  const float* b;           // [D] or null
-  __hip_bfloat16* out;      // head-split [H*B, N, hs]
-  __hip_bfloat16* dz;       // bwd: re-materialized [R, D]
-  __hip_bfloat16* dx;       // bwd: [R, D]
-  float* db;                // bwd: [D] atomics or null
-  long R;                   // B * N
+  const __hip_bfloat16* x;     // [R, D] row-major
+  const float* w;              // [D, D] torch Linear weight (out, in)
+  const float* b;              // [D] or null
+  __hip_bfloat16* out;         // head-split [H*B, N, hs]
+  const __hip_bfloat16* dout;  // bwd in: head-split grad [H*B, N, hs]
+  __hip_bfloat16* dz;          // bwd out: re-materialized [R, D]
+  __hip_bfloat16* dx;          // bwd out: [R, D]
+  float* db;                   // bwd: [D] atomics or null
+  long R;                      // B * N
   int B;
   int N;
 };
@@ -206,8 +207,7 @@ npf_qkv_bwd(QkvParams prm) {
   const int wave = threadIdx.x >> 6;
 
   // dz tile (gathered): logical row-major view of the head-split grad
-  qk_stage_headsplit(pb.dz - 0 + 0 * 0 ? pb.dz : pb.dz, a_lds, r0, pb.R, d,
-                     d_p, pb.B, pb.N, prm.H);
+  qk_stage_headsplit(pb.dout, a_lds, r0, pb.R, d, d_p, pb.B, pb.N, prm.H);
   qk_stage_w(pb.w, w_lds, d, d_p, true);
   __syncthreads();
 
@@ -399,7 +399,8 @@ extern "C" void npf_qkv_fwd_launch(const void* const* xs,
   hipLaunchKernelGGL(npf_qkv_fwd, grid, dim3(QK_BLOCK), 0, stream, p);
 }
 
-extern "C" void npf_qkv_bwd_launch(void* const* dzs, const float* const* ws,
+extern "C" void npf_qkv_bwd_launch(const void* const* douts,
+                                   void* const* dzs, const float* const* ws,
                                    void* const* dxs, float* const* dbs,
                                    const long* Rs, const int* Bs,
                                    const int* Ns, int n_problems, int D,
@@ -407,6 +408,7 @@ extern "C" void npf_qkv_bwd_launch(void* const* dzs, const float* const* ws,
   QkvParams p = {};
   long rmax = 0;
   for (int i = 0; i < n_problems; ++i) {
+    p.p[i].dout = (const __hip_bfloat16*)douts[i];
     p.p[i].dz = (__hip_bfloat16*)dzs[i];
     p.p[i].w = ws[i];
     p.p[i].dx = (__hip_bfloat16*)dxs[i];
@@ -459,8 +461,8 @@ extern "C" void npf_add_ln_bwd_launch(const void* s, const void* dy,
   p.s = (__hip_bfloat16*)s;
   p.dy = (const __hip_bfloat16*)dy;
   p.gamma = gamma;
-  p.mean = mean;
-  p.rstd = rstd;
+  p.mean = (float*)mean;  // read-only in the backward kernel
+  p.rstd = (float*)rstd;
   p.da = (__hip_bfloat16*)da;
   p.y = (__hip_bfloat16*)db;
   p.dgamma = dgamma;

@@ -56,6 +56,19 @@ void npf_lse_z_fwd_launch(const float*, float*, long long, long long,
                           hipStream_t);
 void npf_lse_z_bwd_launch(const float*, const float*, const float*, float*,
                           long long, long long, hipStream_t);
+void npf_qkv_fwd_launch(const void* const*, const float* const*,
+                        const float* const*, void* const*, const long*,
+                        const int*, const int*, int, int, int, hipStream_t);
+void npf_qkv_bwd_launch(const void* const*, void* const*,
+                        const float* const*, void* const*, float* const*,
+                        const long*, const int*, const int*, int, int, int,
+                        hipStream_t);
+void npf_add_ln_fwd_launch(const void*, const void*, const float*,
+                           const float*, void*, void*, float*, float*, long,
+                           int, int, int, int, float, hipStream_t);
+void npf_add_ln_bwd_launch(const void*, const void*, const float*,
+                           const float*, const float*, void*, void*, float*,
+                           float*, long, int, int, int, int, hipStream_t);
 void npf_cb_stats_launch(const void*, int, float*, float*, float*, float*,
                          float*, float*, int, int, int, float, float,
                          hipStream_t);
@@ -291,6 +304,144 @@ torch::Tensor lse_z_bwd(torch::Tensor w, torch::Tensor out,
                        dout.data_ptr<float>(), dw.data_ptr<float>(), Z, B,
                        stream);
   return dw;
+}
+
+// Fused K/Q/V projections with head-split store: for each problem i,
+// out_i = headsplit(x_i @ w_i^T + b_i) with layout [H*B_i, N_i, D/H]
+std::vector<torch::Tensor> qkv_fwd(std::vector<torch::Tensor> xs,
+                                   std::vector<torch::Tensor> ws,
+                                   std::vector<torch::Tensor> bs,
+                                   std::vector<int64_t> Bs,
+                                   std::vector<int64_t> Ns, int64_t H) {
+  const int n = (int)xs.size();
+  TORCH_CHECK(n >= 1 && n <= 3, "qkv: 1..3 problems");
+  const int D = (int)xs[0].size(-1);
+  TORCH_CHECK(D <= 128 && D % (int)(16 * H) == 0,
+              "qkv: D<=128 and head size multiple of 16");
+  const void* xp[3] = {};
+  const float* wp[3] = {};
+  const float* bp[3] = {};
+  void* op[3] = {};
+  long Rs[3] = {};
+  int Bi[3] = {}, Ni[3] = {};
+  std::vector<torch::Tensor> outs;
+  for (int i = 0; i < n; ++i) {
+    check_cuda_contig(xs[i], "qkv x");
+    check_cuda_contig(ws[i], "qkv w");
+    TORCH_CHECK(xs[i].scalar_type() == torch::kBFloat16, "qkv x bf16");
+    TORCH_CHECK(ws[i].scalar_type() == torch::kFloat32, "qkv w fp32");
+    TORCH_CHECK(ws[i].size(0) == D && ws[i].size(1) == D, "qkv square W");
+    TORCH_CHECK((long)Bs[i] * Ns[i] == xs[i].numel() / D, "qkv B*N");
+    auto out = torch::empty({H * Bs[i], Ns[i], D / H},
+                            xs[i].options());
+    outs.push_back(out);
+    xp[i] = xs[i].data_ptr();
+    wp[i] = ws[i].data_ptr<float>();
+    bp[i] = (bs[i].defined() && bs[i].numel() > 0) ? bs[i].data_ptr<float>()
+                                                   : nullptr;
+    op[i] = out.data_ptr();
+    Rs[i] = (long)Bs[i] * Ns[i];
+    Bi[i] = (int)Bs[i];
+    Ni[i] = (int)Ns[i];
+  }
+  auto stream = at::hip::getCurrentHIPStream();
+  npf_qkv_fwd_launch(xp, wp, bp, op, Rs, Bi, Ni, n, D, (int)H, stream);
+  return outs;
+}
+
+// backward: returns per-problem (dx [R,D] bf16, dz [R,D] bf16 row-major)
+// plus db fp32 for problems flagged with bias; the dW GEMMs run in python
+std::vector<torch::Tensor> qkv_bwd(std::vector<torch::Tensor> douts,
+                                   std::vector<torch::Tensor> ws,
+                                   std::vector<int64_t> Bs,
+                                   std::vector<int64_t> Ns,
+                                   std::vector<int64_t> has_bias, int64_t H) {
+  const int n = (int)douts.size();
+  const int D = (int)ws[0].size(0);
+  const void* dop[3] = {};
+  void* dzp[3] = {};
+  const float* wp[3] = {};
+  void* dxp[3] = {};
+  float* dbp[3] = {};
+  long Rs[3] = {};
+  int Bi[3] = {}, Ni[3] = {};
+  std::vector<torch::Tensor> rets;
+  for (int i = 0; i < n; ++i) {
+    check_cuda_contig(douts[i], "qkv dout");
+    const long R = (long)Bs[i] * Ns[i];
+    auto dx = torch::empty({R, (long)D}, douts[i].options());
+    auto dz = torch::empty({R, (long)D}, douts[i].options());
+    torch::Tensor db;
+    if (has_bias[i]) {
+      db = torch::zeros({(long)D}, ws[i].options());
+      dbp[i] = db.data_ptr<float>();
+    }
+    rets.push_back(dx);
+    rets.push_back(dz);
+    rets.push_back(db);
+    dop[i] = douts[i].data_ptr();
+    dzp[i] = dz.data_ptr();
+    wp[i] = ws[i].data_ptr<float>();
+    dxp[i] = dx.data_ptr();
+    Rs[i] = R;
+    Bi[i] = (int)Bs[i];
+    Ni[i] = (int)Ns[i];
+  }
+  auto stream = at::hip::getCurrentHIPStream();
+  npf_qkv_bwd_launch(dop, dzp, wp, dxp, dbp, Rs, Bi, Ni, n, D, (int)H,
+                     stream);
+  return rets;
+}
+
+// y = LayerNorm(a + b); a optionally in head-split layout (H != 0)
+std::vector<torch::Tensor> add_ln_fwd(torch::Tensor a, torch::Tensor b,
+                                      torch::Tensor gamma, torch::Tensor beta,
+                                      int64_t B, int64_t N, int64_t H,
+                                      double eps) {
+  check_cuda_contig(a, "aln a");
+  check_cuda_contig(b, "aln b");
+  TORCH_CHECK(a.scalar_type() == torch::kBFloat16 &&
+                  b.scalar_type() == torch::kBFloat16,
+              "add_ln bf16 inputs");
+  const int D = (int)b.size(-1);
+  TORCH_CHECK(D <= 128, "add_ln D<=128");
+  const long R = b.numel() / D;
+  auto y = torch::empty_like(b.reshape({R, (long)D}));
+  auto s = torch::empty_like(y);
+  auto mean = torch::empty({R}, gamma.options());
+  auto rstd = torch::empty({R}, gamma.options());
+  auto stream = at::hip::getCurrentHIPStream();
+  npf_add_ln_fwd_launch(a.data_ptr(), b.data_ptr(),
+                        gamma.data_ptr<float>(), beta.data_ptr<float>(),
+                        y.data_ptr(), s.data_ptr(), mean.data_ptr<float>(),
+                        rstd.data_ptr<float>(), R, D, (int)B, (int)N, (int)H,
+                        (float)eps, stream);
+  return {y, s, mean, rstd};
+}
+
+std::vector<torch::Tensor> add_ln_bwd(torch::Tensor s, torch::Tensor dy,
+                                      torch::Tensor gamma, torch::Tensor mean,
+                                      torch::Tensor rstd, int64_t B,
+                                      int64_t N, int64_t H) {
+  check_cuda_contig(s, "aln s");
+  check_cuda_contig(dy, "aln dy");
+  const int D = (int)s.size(-1);
+  const long R = s.numel() / D;
+  torch::Tensor da;
+  if (H != 0)
+    da = torch::empty({H * B, N, (long)(D / H)}, s.options());
+  else
+    da = torch::empty_like(s);
+  auto db = torch::empty_like(s);
+  auto dgamma = torch::zeros_like(gamma);
+  auto dbeta = torch::zeros_like(gamma);
+  auto stream = at::hip::getCurrentHIPStream();
+  npf_add_ln_bwd_launch(s.data_ptr(), dy.data_ptr(),
+                        gamma.data_ptr<float>(), mean.data_ptr<float>(),
+                        rstd.data_ptr<float>(), da.data_ptr(), db.data_ptr(),
+                        dgamma.data_ptr<float>(), dbeta.data_ptr<float>(), R,
+                        D, (int)B, (int)N, (int)H, stream);
+  return {da, db, dgamma, dbeta};
 }
 
 // fused conv block: stats (training BN) -> (mean, rstd, save_var)
@@ -612,6 +763,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gauss_ll_bwd", &gauss_ll_bwd, "fused Gaussian log-lik backward");
   m.def("lse_z_fwd", &lse_z_fwd, "NPML logmeanexp-over-z forward");
   m.def("lse_z_bwd", &lse_z_bwd, "NPML logmeanexp-over-z backward");
+  m.def("qkv_fwd", &qkv_fwd, "fused K/Q/V projections + head split");
+  m.def("qkv_bwd", &qkv_bwd, "fused K/Q/V projection backward");
+  m.def("add_ln_fwd", &add_ln_fwd, "fused add + LayerNorm forward");
+  m.def("add_ln_bwd", &add_ln_bwd, "fused add + LayerNorm backward");
   m.def("convblock_stats", &convblock_stats,
         "fused conv block: per-channel batch stats (+running update)");
   m.def("convblock_fwd", &convblock_fwd,

@@ -335,8 +335,74 @@ class TransformerAttender(MultiheadAttender):
         )
         self.reset_parameters()
 
+    def _fused_block_ok(self, keys, queries, values):
+        from npf.ops import has_extension
+
+        return (
+            keys.is_cuda
+            and has_extension()
+            and not self.is_relative_pos
+            and self.kq_size == self.value_size == self.out_size
+            and self.kq_size <= 128
+            and self.kq_head_size == 16
+            and isinstance(self.dot.dropout, nn.Identity)
+            and keys.shape[-1] == self.kq_size
+            and self.key_transform.weight.dtype == torch.float32
+        )
+
     def forward(self, keys, queries, values, **kwargs):
+        if self._fused_block_ok(keys, queries, values):
+            return self._fused_forward(keys, queries, values)
         context = super().forward(keys, queries, values, **kwargs)
         context = self.layer_norm1(context + queries)
         context = self.layer_norm2(context + self.dot.dropout(self.mlp(context)))
         return context
+
+    def _fused_forward(self, keys, queries, values):
+        """5 kernels for the whole block (vs ~18 + casts composed):
+        qkv-projection+head-split MFMA kernel -> fused attention ->
+        add+LN1 (head merge folded into the gather) -> mlp_chain FFN ->
+        add+LN2.  Numerics = autocast bf16 with fp32 statistics."""
+        from npf.ops import attention_qkv
+        from npf.ops.functional import add_layernorm, qkv_project_headsplit
+
+        B, Qn, D = queries.shape
+        Kn = keys.shape[1]
+        if keys is queries and queries is values:
+            (kh,) = qkv_project_headsplit(
+                [keys], [self.key_transform.weight], [None], self.n_heads
+            )
+            qh, vh = qkv_project_headsplit(
+                [queries, values],
+                [self.query_transform.weight, self.value_transform.weight],
+                [self.query_transform.bias, None],
+                self.n_heads,
+            )
+        else:
+            kh, qh, vh = qkv_project_headsplit(
+                [keys, queries, values],
+                [
+                    self.key_transform.weight,
+                    self.query_transform.weight,
+                    self.value_transform.weight,
+                ],
+                [None, self.query_transform.bias, None],
+                self.n_heads,
+            )
+        ctx_h = attention_qkv(kh, qh, vh)  # [H*B, Qn, 16]
+        h1 = add_layernorm(
+            ctx_h,
+            queries,
+            self.layer_norm1.weight,
+            self.layer_norm1.bias,
+            eps=self.layer_norm1.eps,
+            headsplit=(B, Qn, self.n_heads),
+        ).view(B, Qn, D)
+        ffn = self.mlp(h1)  # mlp_chain fused on GPU
+        return add_layernorm(
+            ffn.view(B, Qn, D),
+            h1,
+            self.layer_norm2.weight,
+            self.layer_norm2.bias,
+            eps=self.layer_norm2.eps,
+        ).view(B, Qn, D)

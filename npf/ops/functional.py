@@ -299,6 +299,103 @@ def gaussian_nll_logmeanexp(loc, scale, y):
 
 
 # --------------------------------------------------------------------------- #
+# Fused TransformerAttender pieces (reference attention.py:375-527, :530-588).
+# qkv_project_headsplit: the K/Q/V projection Linears as one MFMA kernel that
+# writes the head-split [H*B, N, hs] layout directly (no permute kernels).
+# add_layernorm: y = LayerNorm(a + b) with an optional head-split gather on
+# `a`, so the attention output never needs a head-merge kernel either.
+# --------------------------------------------------------------------------- #
+
+
+class _QkvProjFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, n_heads, Bs, Ns, n_problems, *tensors):
+        ext = _backend.require_extension("qkv_project_headsplit")
+        xs = list(tensors[:n_problems])
+        ws = list(tensors[n_problems : 2 * n_problems])
+        bs = list(tensors[2 * n_problems :])
+        outs = ext.qkv_fwd(
+            xs, ws, [b if b is not None else torch.Tensor() for b in bs],
+            list(Bs), list(Ns), n_heads,
+        )
+        ctx.save_for_backward(*xs, *ws)
+        ctx.meta = (n_heads, tuple(Bs), tuple(Ns), n_problems,
+                    tuple(b is not None for b in bs))
+        return tuple(outs)
+
+    @staticmethod
+    def backward(ctx, *douts):
+        ext = _backend.extension()
+        n_heads, Bs, Ns, n, has_bias = ctx.meta
+        xs = ctx.saved_tensors[:n]
+        ws = ctx.saved_tensors[n:]
+        rets = ext.qkv_bwd(
+            [d.contiguous() for d in douts], list(ws), list(Bs), list(Ns),
+            [int(h) for h in has_bias], n_heads,
+        )
+        dxs, dws, dbs = [], [], []
+        for i in range(n):
+            dx, dz, db = rets[3 * i], rets[3 * i + 1], rets[3 * i + 2]
+            dxs.append(dx.view_as(xs[i]))
+            # dW = dz^T @ x : K-large GEMM on hipBLASLt, fp32 master grad
+            dws.append(torch.mm(dz.t(), xs[i].reshape(-1, xs[i].shape[-1])).float())
+            dbs.append(db if has_bias[i] else None)
+        return (None, None, None, None, *dxs, *dws, *dbs)
+
+
+def qkv_project_headsplit(xs, weights, biases, n_heads):
+    """[(B, N, D)] inputs -> [(H*B, N, D/H)] projected head-split outputs.
+
+    One MFMA kernel for up to 3 projection problems (K, Q, V); torch Linear
+    weight convention ([out, in]); bias may be None per problem.
+    """
+    Bs = [x.shape[0] for x in xs]
+    Ns = [x.shape[1] for x in xs]
+    xs = [x.reshape(-1, x.shape[-1]).to(torch.bfloat16).contiguous() for x in xs]
+    ws = [w.float() for w in weights]
+    bs = [b.float() if b is not None else None for b in biases]
+    return _QkvProjFn.apply(n_heads, Bs, Ns, len(xs), *xs, *ws, *bs)
+
+
+class _AddLNFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, a, b, gamma, beta, B, N, H, eps):
+        ext = _backend.require_extension("add_layernorm")
+        y, s, mean, rstd = ext.add_ln_fwd(a, b, gamma, beta, B, N, H, eps)
+        ctx.save_for_backward(s, gamma, mean, rstd)
+        ctx.meta = (B, N, H)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        ext = _backend.extension()
+        s, gamma, mean, rstd = ctx.saved_tensors
+        B, N, H = ctx.meta
+        da, db, dgamma, dbeta = ext.add_ln_bwd(
+            s, dy.reshape(s.shape).contiguous(), gamma, mean, rstd, B, N, H
+        )
+        return da, db, dgamma, dbeta, None, None, None, None
+
+
+def add_layernorm(a, b, gamma, beta, eps=1e-5, headsplit=None):
+    """LayerNorm(a + b) over the last dim (<= 128), bf16 in/out, fp32 stats.
+
+    `headsplit=(B, N, H)` reads `a` in the attention head-split layout
+    [H*B, N, D/H] (row r of the plain view maps to (b=r//N, n=r%N)); the
+    returned grad for `a` keeps that layout.  b: [B*N, D] or [B, N, D].
+    """
+    B, N, H = headsplit if headsplit is not None else (0, 0, 0)
+    lead = b.shape[:-1]
+    D = b.shape[-1]
+    y = _AddLNFn.apply(
+        a.to(torch.bfloat16).contiguous(),
+        b.reshape(-1, D).to(torch.bfloat16).contiguous(),
+        gamma.float(), beta.float(), B, N, H, eps,
+    )
+    return y.reshape(*lead, D)
+
+
+# --------------------------------------------------------------------------- #
 # Fused pre-activation depthwise conv block (1D).
 # Reference computation: npf/architectures/cnn.py ResConvBlock.forward
 # (reference cnn.py:204-215) — batchnorm -> relu -> depthwise conv1d
