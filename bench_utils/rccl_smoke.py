@@ -11,6 +11,8 @@ Two ranks on one GPU is 'invalid usage' for RCCL (duplicate-device), so the
 import os
 import sys
 
+sys.path.insert(0, "/root/repo")
+
 import torch
 import torch.distributed as dist
 

@@ -167,22 +167,22 @@ npf_cb2d_bwd_dact(const T* __restrict__ x, const float* __restrict__ w,
   }
   __syncthreads();
   // dW[kr,kc] = sum_p dY[p] * a[p + (kr,kc)]: K^2 positions split across
-  // waves, shuffle-only reduction, one atomic per (row-tile, position)
+  // waves, shuffle-only reduction, one atomic per (row-tile, position).
+  // Row/col loops instead of a strided pixel walk: the inner loop is pure
+  // lane-strided adds (the walk's per-element index fixups were measured
+  // as several x the stencil VALU work in PMC counters)
   {
     const int lane = threadIdx.x & 63;
     const int wv = threadIdx.x >> 6;
     const int nw = CB2_BLOCK / 64;
-    const int r_in = lane / W, c_in = lane % W;
-    const int drw = 64 / W, dcw = 64 % W;
     for (int kk = wv; kk < K * K; kk += nw) {
       const int kr = kk / K, kc = kk % K;
       float psum = 0.f;
-      int r = r_in, col = c_in;
-      while (r < tr) {
-        psum += dys[(r + pad) * WP + col + pad] * a[(r + kr) * WP + col + kc];
-        r += drw;
-        col += dcw;
-        if (col >= W) { col -= W; ++r; }
+      for (int r = 0; r < tr; ++r) {
+        const float* dyrow = dys + (r + pad) * WP + pad;
+        const float* arow = a + (r + kr) * WP + kc;
+        for (int col = lane; col < W; col += 64)
+          psum += dyrow[col] * arow[col];
       }
       psum = wave_reduce_sum(psum);
       if (lane == 0) atomicAdd(&dw[c * K * K + kk], psum);

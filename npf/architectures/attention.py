@@ -336,11 +336,14 @@ class TransformerAttender(MultiheadAttender):
         self.reset_parameters()
 
     def _fused_block_ok(self, keys, queries, values):
+        import os
+
         from npf.ops import has_extension
 
         return (
             keys.is_cuda
             and has_extension()
+            and os.environ.get("NPF_FORCE_EAGER") != "1"
             and not self.is_relative_pos
             and self.kq_size == self.value_size == self.out_size
             and self.kq_size <= 128

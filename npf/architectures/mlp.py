@@ -7,6 +7,7 @@ hipBLASLt bf16 GEMMs under autocast; the (Linear -> activation) chains are
 fusion candidates for the HIP epilogue kernels (SURVEY.md §2.3 "MLP decoder").
 """
 
+import os
 import warnings
 
 import torch
@@ -85,6 +86,7 @@ class MLP(nn.Module):
             and self.to_hidden.bias is not None
             and self.to_hidden.weight.dtype == torch.float32
             and has_extension()
+            and os.environ.get("NPF_FORCE_EAGER") != "1"
         )
 
     def forward(self, x):
