@@ -17,11 +17,6 @@ trainer shards tasks per rank and all-reduces gradients over RCCL/xGMI.
 """
 
 import argparse
-import os
-import sys
-
-# the canonical model builders live next to the tests so they are shared by
-# tests, benches and this CLI without duplicating configs
 
 MODELS_1D = {
     "CNP": "cnp_1d",
@@ -95,6 +90,22 @@ def _build_data(args):
     return train, test, dict(), False
 
 
+def _raw_splitter(is_1d):
+    from npf.utils.datasplit import (
+        CntxtTrgtGetter, GetRandomIndcs, GridCntxtTrgtGetter, RandomMasker,
+        get_all_indcs, no_masker,
+    )
+
+    if is_1d:
+        return CntxtTrgtGetter(
+            contexts_getter=GetRandomIndcs(a=0.0, b=50),
+            targets_getter=get_all_indcs,
+        )
+    return GridCntxtTrgtGetter(
+        context_masker=RandomMasker(a=0.0, b=0.3), target_masker=no_masker
+    )
+
+
 def _splitter(is_1d, img_shape=None):
     from npf.data.dataloader import cntxt_trgt_collate
     from npf.utils.datasplit import (
@@ -139,6 +150,16 @@ def _common(args, is_retrain):
         y_dim = next(iter(train.values())).shape[0]
         builder = partial(builder, y_dim=y_dim)
 
+    # loss-ablation knobs (the 24-model grid of the reference Losses.ipynb):
+    # lower bounds on predictive / latent std via the transformer kwargs
+    if args.min_sigma_pred != 0.01 or args.min_lat is not None:
+        from npf.recipes import get_std_processing_kwargs
+
+        std_kwargs = get_std_processing_kwargs(
+            min_sigma_pred=args.min_sigma_pred, min_lat=args.min_lat
+        )
+        builder = partial(_apply_std_kwargs, builder, std_kwargs)
+
     criterion = _build_loss(args.model, args.loss)
     collate = _splitter(is_1d)
 
@@ -162,8 +183,23 @@ def _common(args, is_retrain):
         iterator_train__collate_fn=collate,
         iterator_valid__collate_fn=collate,
         amp_dtype="bfloat16" if args.bf16 else None,
+        device_episodes=_raw_splitter(is_1d) if args.device_episodes else None,
+        hipgraphs=args.hipgraphs,
     )
     return trainers
+
+
+def _apply_std_kwargs(builder, std_kwargs):
+    model = builder()
+    import torch.nn as nn
+
+    if "p_y_scale_transformer" in std_kwargs:
+        model.p_y_scale_transformer = std_kwargs["p_y_scale_transformer"]
+    if "q_z_scale_transformer" in std_kwargs and hasattr(
+        model, "q_z_scale_transformer"
+    ):
+        model.q_z_scale_transformer = std_kwargs["q_z_scale_transformer"]
+    return model
 
 
 def cmd_train(args):
@@ -198,6 +234,14 @@ def main(argv=None):
         q.add_argument("--data-cache", default="data/gp_dataset.npz")
         q.add_argument("--device", default=None)
         q.add_argument("--bf16", action="store_true")
+        q.add_argument("--device-episodes", action="store_true",
+                       help="GPU-resident episodes (no DataLoader/collate)")
+        q.add_argument("--hipgraphs", action="store_true",
+                       help="capture optimization steps per episode shape")
+        q.add_argument("--min-sigma-pred", type=float, default=0.01,
+                       help="predictive-std lower bound (ablation grid knob)")
+        q.add_argument("--min-lat", type=float, default=None,
+                       help="latent-std lower bound (ablation grid knob)")
         q.set_defaults(fn=fn)
 
     args = p.parse_args(argv)

@@ -158,3 +158,24 @@ def test_std_processing_kwargs_ablation_grid():
             X = torch.rand(2, 9, 1) * 2 - 1
             p, *_ = m(X[:, :4], torch.randn(2, 4, 1), X)
             assert float(p.base_dist.scale.detach().min()) >= min_sig
+
+
+def test_cli_ablation_knobs_apply(tmp_path):
+    """--min-sigma-pred / --min-lat (the Losses.ipynb grid knobs) change the
+    scale transformers and the model still trains a step."""
+    import numpy as np
+
+    from npf import cli
+
+    cli.main([
+        "train", "--model", "LNP", "--data", "RBF_Kernel",
+        "--epochs", "1", "--n-tasks", "64", "--n-test-tasks", "16",
+        "--batch-size", "8", "--chckpnt-dir", str(tmp_path) + "/",
+        "--data-cache", str(tmp_path / "cache.npz"),
+        "--min-sigma-pred", "0.1", "--min-lat", "0.1",
+        "--loss", "elbo",
+    ])
+    run_dir = tmp_path / "RBF_Kernel" / "LNP" / "run_0"
+    assert (run_dir / "params.pt").exists()
+    ll = np.loadtxt(run_dir / "eval.csv", delimiter=",")
+    assert ll.shape == (16,) and np.isfinite(ll).all()
