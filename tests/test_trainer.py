@@ -365,3 +365,39 @@ class TestGraphedStepEquivalence:
             assert torch.allclose(p1, p2, atol=5e-2), (
                 n1, (p1 - p2).abs().max()
             )
+
+
+def test_device_episodes_image_grid():
+    """Grid (image) datasets run through the device-resident loader: masker
+    splitting on-device, GridConvCNP steps."""
+    from npf import CNPFLoss
+    from npf.data.imgs import SyntheticImages
+    from npf.train.device_loader import DeviceEpisodes
+    from npf.train.trainer import NPFTrainer
+    from npf.utils.datasplit import GridCntxtTrgtGetter, RandomMasker, no_masker
+    from npf.zoo import gridconvcnp_2d
+
+    from functools import partial
+
+    set_seed(0)
+    ds = SyntheticImages(shape=(3, 16, 16), n_samples=16)
+    eps = DeviceEpisodes(
+        ds,
+        partial(
+            GridCntxtTrgtGetter(
+                context_masker=RandomMasker(a=0.05, b=0.2),
+                target_masker=no_masker,
+            ),
+            is_return_masks=True,  # GridConv models consume mask episodes
+        ),
+        device="cpu",
+    )
+    tr = NPFTrainer(
+        gridconvcnp_2d(y_dim=3), CNPFLoss(), device="cpu", batch_size=8,
+        max_epochs=1, seed=0,
+    )
+    tr.fit(eps)
+    assert tr.history[0]["train_loss"] is not None
+    import math
+
+    assert math.isfinite(tr.history[0]["train_loss"])
