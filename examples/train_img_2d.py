@@ -33,6 +33,9 @@ def main():
     p.add_argument("--epochs", type=int, default=50)
     p.add_argument("--chckpnt-dir", default="results/pretrained/")
     p.add_argument("--bf16", action="store_true")
+    p.add_argument("--device-episodes", action="store_true")
+    p.add_argument("--hipgraphs", action="store_true")
+    p.add_argument("--batch-size", type=int, default=32)
     args = p.parse_args()
 
     train, test = get_img_datasets(args.datasets)
@@ -41,13 +44,14 @@ def main():
         builder, criterion, is_grid = MODELS[name]
         models = add_y_dim({name: builder}, train)
         # reference 2D splitter: U(0, 30%) of pixels as context
-        collate = cntxt_trgt_collate(
-            GridCntxtTrgtGetter(
-                context_masker=RandomMasker(a=0.0, b=0.3),
-                target_masker=no_masker,
-            ),
-            is_return_masks=is_grid,
+        splitter = GridCntxtTrgtGetter(
+            context_masker=RandomMasker(a=0.0, b=0.3),
+            target_masker=no_masker,
         )
+        collate = cntxt_trgt_collate(splitter, is_return_masks=is_grid)
+        from functools import partial
+
+        dev_split = partial(splitter, is_return_masks=is_grid)
         train_models(
             train,
             models,
@@ -57,13 +61,16 @@ def main():
             is_retrain=True,
             train_split=CVSplit(0.1),
             max_epochs=args.epochs,
-            batch_size=32,
+            batch_size=args.batch_size,
             lr=1e-3,
             decay_lr=10,
             seed=123,
             iterator_train__collate_fn=collate,
             iterator_valid__collate_fn=collate,
             amp_dtype="bfloat16" if args.bf16 else None,
+            device_episodes=dev_split if args.device_episodes else None,
+            hipgraphs=args.hipgraphs,
+            is_progressbar=True,
         )
 
 

@@ -590,14 +590,19 @@ def train_models(
                     trainer.load_params()
 
                 if is_retrain:
-                    fit_data = data_train
+                    fit_data, fit_valid = data_train, None
                     if device_episodes is not None:
                         from npf.train.device_loader import DeviceEpisodes
 
+                        # split BEFORE wrapping (Subset of the raw dataset);
+                        # validation keeps the DataLoader/collate path
+                        if trainer.train_split is not None:
+                            fit_data, fit_valid = trainer.train_split(fit_data)
+                            trainer.train_split = None
                         fit_data = DeviceEpisodes(
-                            data_train, device_episodes, device=trainer.device
+                            fit_data, device_episodes, device=trainer.device
                         )
-                    trainer.fit(fit_data)
+                    trainer.fit(fit_data, valid_dataset=fit_valid)
                     if run_dir and dist_utils.get_rank() == 0:
                         with open(os.path.join(run_dir, MOD_SUMM_FILENAME), "w") as f:
                             f.write(str(trainer.module))
