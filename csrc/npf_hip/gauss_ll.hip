@@ -72,10 +72,10 @@ npf_lse_z_fwd(const float* __restrict__ w, float* __restrict__ out,
   mx = __shfl(mx, 0);
   float s = 0.f;
   for (long long z = threadIdx.x; z < Z; z += 64)
-    s += __expf(w[z * B + b] - mx);
+    s += expf(w[z * B + b] - mx);
   for (int off = 32; off; off >>= 1)
     s += __shfl_down(s, off);
-  if (threadIdx.x == 0) out[b] = mx + __logf(s) - __logf((float)Z);
+  if (threadIdx.x == 0) out[b] = mx + logf(s) - logf((float)Z);
 }
 
 // dw[z,b] = dout[b] * softmax_z(w)[z,b] = dout[b] * exp(w - (out[b] + log Z))
@@ -86,8 +86,8 @@ npf_lse_z_bwd(const float* __restrict__ w, const float* __restrict__ out,
   const long long i = (long long)blockIdx.x * 256 + threadIdx.x;
   if (i >= Z * B) return;
   const long long b = i % B;
-  const float lse = out[b] + __logf((float)Z);
-  dw[i] = dout[b] * __expf(w[i] - lse);
+  const float lse = out[b] + logf((float)Z);
+  dw[i] = dout[b] * expf(w[i] - lse);
 }
 
 extern "C" void npf_lse_z_fwd_launch(const float* w, float* out, long long Z,

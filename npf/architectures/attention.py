@@ -402,10 +402,15 @@ class TransformerAttender(MultiheadAttender):
             headsplit=(B, Qn, self.n_heads),
         ).view(B, Qn, D)
         ffn = self.mlp(h1)  # mlp_chain fused on GPU
-        return add_layernorm(
+        out = add_layernorm(
             ffn.view(B, Qn, D),
             h1,
             self.layer_norm2.weight,
             self.layer_norm2.bias,
             eps=self.layer_norm2.eps,
         ).view(B, Qn, D)
+        if not torch.is_autocast_enabled("cuda"):
+            # outside autocast the caller's modules are fp32: hand back the
+            # caller's dtype (under autocast bf16 flows on unchanged)
+            out = out.to(queries.dtype)
+        return out

@@ -76,7 +76,9 @@ class MLP(nn.Module):
 
         return (
             x.is_cuda
-            and torch.is_autocast_enabled("cuda")
+            # bf16 inputs arrive from other fused ops (e.g. the fused
+            # attender) outside autocast: same numerics, same kernel
+            and (torch.is_autocast_enabled("cuda") or x.dtype == torch.bfloat16)
             and isinstance(self.activation, nn.ReLU)
             and isinstance(self.dropout, nn.Identity)
             and not self.is_res

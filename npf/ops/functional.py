@@ -387,6 +387,9 @@ def add_layernorm(a, b, gamma, beta, eps=1e-5, headsplit=None):
     B, N, H = headsplit if headsplit is not None else (0, 0, 0)
     lead = b.shape[:-1]
     D = b.shape[-1]
+    if H == 0:
+        # plain layout: flatten so the returned grad matches a's shape
+        a = a.reshape(-1, D)
     y = _AddLNFn.apply(
         a.to(torch.bfloat16).contiguous(),
         b.reshape(-1, D).to(torch.bfloat16).contiguous(),

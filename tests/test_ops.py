@@ -682,11 +682,11 @@ class TestNLLLogMeanExpGPU:
         scale0 = scale.detach().cpu().requires_grad_()
         out0 = gaussian_nll_logmeanexp(loc0, scale0, y.cpu())
         out0.sum().backward()
-        assert torch.allclose(out.cpu(), out0, atol=1e-4), (
+        assert torch.allclose(out.cpu(), out0, atol=1e-3), (
             (out.cpu() - out0).abs().max()
         )
-        assert torch.allclose(loc.grad.cpu(), loc0.grad, atol=1e-4)
-        assert torch.allclose(scale.grad.cpu(), scale0.grad, atol=1e-4)
+        assert torch.allclose(loc.grad.cpu(), loc0.grad, atol=1e-3)
+        assert torch.allclose(scale.grad.cpu(), scale0.grad, atol=1e-3)
 
 
 @pytest.mark.gpu
