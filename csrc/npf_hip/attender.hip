@@ -363,11 +363,29 @@ npf_add_ln_bwd(AddLnParams p) {
       const __hip_bfloat16 h = __float2bfloat16(ds);
       p.da[aln_a_index(p, r, c)] = h;       // grad w.r.t. a (a's layout)
       p.y[r * D + c] = h;                   // y reused as db (plain layout)
-      const float dyv = __bfloat162float(p.dy[r * D + c]);
-      atomicAdd(&p.dgamma[c], dyv * xh[i]);
-      atomicAdd(&p.dbeta[c], dyv);
     }
   }
+}
+
+// dgamma[c] = sum_r dy[r,c] * xhat[r,c]; dbeta[c] = sum_r dy[r,c].
+// Separate column-reduction pass: per-element atomics in the row kernel
+// were 524k global atomics on 128 addresses.  Each thread owns a column
+// (coalesced across the warp), each workgroup strides a row slab; one
+// atomic per (workgroup, column).
+extern "C" __global__ void __launch_bounds__(128)
+npf_add_ln_gb(AddLnParams p) {
+  const int c = threadIdx.x;
+  if (c >= p.D) return;
+  float sg = 0.f, sb = 0.f;
+  for (long r = blockIdx.x; r < p.R; r += gridDim.x) {
+    const float dyv = __bfloat162float(p.dy[r * p.D + c]);
+    const float xh =
+        (__bfloat162float(p.s[r * p.D + c]) - p.mean[r]) * p.rstd[r];
+    sg += dyv * xh;
+    sb += dyv;
+  }
+  atomicAdd(&p.dgamma[c], sg);
+  atomicAdd(&p.dbeta[c], sb);
 }
 
 // ---------------------------------------------------------------------------
@@ -475,4 +493,6 @@ extern "C" void npf_add_ln_bwd_launch(const void* s, const void* dy,
   const unsigned grid = (unsigned)((R + ALN_WAVES - 1) / ALN_WAVES);
   hipLaunchKernelGGL(npf_add_ln_bwd, dim3(grid), dim3(64 * ALN_WAVES), 0,
                      stream, p);
+  const unsigned gb_grid = (unsigned)(R < 512 ? ((R + 7) / 8 + 1) : 64);
+  hipLaunchKernelGGL(npf_add_ln_gb, dim3(gb_grid), dim3(128), 0, stream, p);
 }
