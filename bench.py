@@ -176,6 +176,9 @@ def _configs():
             needs_y_trgt=True,  # NPVI: q_zCct from the target set
             is_latent=True,  # rsample routed through the static noise pool
                              # (npf.ops.noise) so the step captures cleanly
+            tune_ok=False,   # TunableOp's hipBLASLt sweep on these GEMM
+                             # shapes memory-faults the box (observed twice
+                             # on fresh leases, 2026-09-13); skip tuning
             desc="AttnLNP-2D CelebA32-shape (self-attn encoder, NPVI, 468,486 params)",
             seq_len=32 * 32,
             n_cntxt=int(0.3 * 32 * 32),
@@ -223,7 +226,7 @@ def main():
         assert world == args.gpus or args.gpus == 1, (world, args.gpus)
     n_gpus = world if world > 1 else 1
 
-    if use_cuda and not os.environ.get("NPF_BENCH_NO_TUNE"):
+    if use_cuda and not os.environ.get("NPF_BENCH_NO_TUNE") and cfg.get("tune_ok", True):
         # per-shape GEMM autotuning (rocBLAS/hipBLASLt/CK): tunes during
         # warmup, then the tuned kernels are what the graph captures.
         # NPF_BENCH_NO_TUNE=1 skips it (profiling runs: tuning floods the
