@@ -734,16 +734,20 @@ class TestFusedTransformerAttender:
         keys = torch.randn(2, 13, 128, generator=g)
         queries = torch.randn(2, 37, 128, generator=g)
         values = torch.randn(2, 13, 128, generator=g)
+        # NOT square().sum(): that loss is numerically invariant to a
+        # LayerNorm input (sum xhat^2 == D per row), so true input grads
+        # are ~0 and bf16 noise dominates any relative comparison
+        w = torch.randn(2, 37, 128, generator=g)
 
         kc = keys.cuda().requires_grad_()
         qc = queries.cuda().requires_grad_()
         vc = values.cuda().requires_grad_()
-        m(kc, qc, vc).square().sum().backward()
+        (m(kc, qc, vc) * w.cuda()).sum().backward()
 
         k0 = keys.clone().requires_grad_()
         q0 = queries.clone().requires_grad_()
         v0 = values.clone().requires_grad_()
-        m0(k0, q0, v0).square().sum().backward()
+        (m0(k0, q0, v0) * w).sum().backward()
 
         for a, b, name in [
             (kc.grad, k0.grad, "dk"), (qc.grad, q0.grad, "dq"),
