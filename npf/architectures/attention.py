@@ -344,6 +344,11 @@ class TransformerAttender(MultiheadAttender):
             keys.is_cuda
             and has_extension()
             and os.environ.get("NPF_FORCE_EAGER") != "1"
+            # bf16 block: engage only under autocast / bf16 inputs — fp32
+            # training must keep fp32 numerics (the LL targets live in a
+            # regime where bf16 rounding is irreducible noise)
+            and (torch.is_autocast_enabled("cuda")
+                 or keys.dtype == torch.bfloat16)
             and not self.is_relative_pos
             and self.kq_size == self.value_size == self.out_size
             and self.kq_size <= 128
