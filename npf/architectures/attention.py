@@ -344,6 +344,7 @@ class TransformerAttender(MultiheadAttender):
             keys.is_cuda
             and has_extension()
             and os.environ.get("NPF_FORCE_EAGER") != "1"
+            and os.environ.get("NPF_NO_FUSED_ATTENDER") != "1"
             # bf16 block: engage only under autocast / bf16 inputs — fp32
             # training must keep fp32 numerics (the LL targets live in a
             # regime where bf16 rounding is irreducible noise)
