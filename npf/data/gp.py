@@ -192,7 +192,11 @@ class GPDataset(Dataset):
         device = self.device
         n_groups = (n_samples + self.n_same_samples - 1) // self.n_same_samples
 
-        X = torch.empty(n_groups, n_points, 1, device=device)
+        # float64 factorization at the reference's jitter scale (sklearn
+        # GaussianProcessRegressor alpha=1e-10): RBF ls=0.2 over 128 points
+        # is too ill-conditioned for fp32, and a bigger jitter would raise
+        # the data's noise floor above what the published LLs assume
+        X = torch.empty(n_groups, n_points, 1, device=device, dtype=torch.float64)
         X.uniform_(min_max[0], min_max[1])
         X, _ = X.sort(dim=1)
 
@@ -205,8 +209,11 @@ class GPDataset(Dataset):
         else:
             cov = self.kernel(X)
 
-        L = _robust_cholesky(cov)
-        eps = torch.randn(n_groups, n_points, self.n_same_samples, device=device)
+        L = _robust_cholesky(cov.double(), jitter=1e-10)
+        eps = torch.randn(
+            n_groups, n_points, self.n_same_samples, device=device,
+            dtype=torch.float64,
+        )
         # [G, N, S] -> per-group S functions
         Y = torch.bmm(L, eps)
 
