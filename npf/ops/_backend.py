@@ -45,6 +45,8 @@ def require_extension(op_name):
     """Return the extension, raising loudly if a GPU op has no native kernel."""
     if os.environ.get("NPF_FORCE_EAGER") == "1":  # debug: composed-torch path
         return None
+    if op_name in os.environ.get("NPF_DISABLE_OPS", "").split(","):
+        return None  # selective composed fallback (kernel bisection)
     ext = _try_load()
     if ext is None and os.environ.get("NPF_ALLOW_EAGER_GPU") != "1":
         raise RuntimeError(
