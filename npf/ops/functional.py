@@ -406,6 +406,14 @@ def add_layernorm(a, b, gamma, beta, eps=1e-5, headsplit=None):
 # --------------------------------------------------------------------------- #
 
 
+def _bump_num_batches(bn, training):
+    """torch BN increments num_batches_tracked per training forward; the
+    fused kernels update running stats directly, so mirror the counter here
+    (a device-side add: it replays correctly inside captured graphs)."""
+    if training and bn is not None and getattr(bn, "num_batches_tracked", None) is not None:
+        bn.num_batches_tracked.add_(1)
+
+
 def _conv_block_ref(x, weight, bias, bn, residual, training):
     a = x
     if bn is not None:
@@ -508,6 +516,7 @@ def conv_block_1d(x, conv, bn=None, residual=None):
             residual, bn.training if bn is not None else conv.training,
         )
     training = bn.training if bn is not None else conv.training
+    _bump_num_batches(bn, training)
     # fp32 compute: BN statistics and the stencil are precision-sensitive;
     # at these sizes the op is dispatch/HBM-bound so bf16 buys nothing
     xf = x.float().contiguous()
@@ -633,6 +642,7 @@ def conv_block_2d(x, conv, bn=None, residual=None):
         return _conv_block2d_ref(
             x, conv.weight, conv.bias, bn, residual, training
         )
+    _bump_num_batches(bn, training)
     xf = x.contiguous()
     res = residual.to(x.dtype).contiguous() if residual is not None else None
     return _ConvBlock2dFn.apply(
