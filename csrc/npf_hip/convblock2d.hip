@@ -17,12 +17,10 @@
 
 #define CB2_BLOCK 256
 #define CB2_MAX_K 13
-#define CB2_TR 16  // plane rows per workgroup: small LDS tiles keep enough
+#define CB2_TR 8   // plane rows per workgroup: small LDS tiles -> enough
                    // resident waves to hide LDS latency (a full 64x64 plane
-                   // per WG was 41.5 KB -> 3 waves/SIMD -> 8.7ms bwd);
-                   // 16 rows halve the k/2-halo staging overhead vs 8 and
-                   // with 4-column register blocking 16x16 quads = exactly
-                   // one position per thread at W=64
+                   // per WG was 41.5 KB -> 3 waves/SIMD -> 8.7ms bwd; row
+                   // tiles re-read the k/2 halo rows but run latency-hidden)
 
 template <typename T, int KT>
 __global__ void __launch_bounds__(CB2_BLOCK)
@@ -63,42 +61,25 @@ npf_cb2d_fwd(const T* __restrict__ x, const T* __restrict__ res,
   T* ypl = y + ((long)n * C + c) * H * W;
   const T* rpl =
       (res != nullptr) ? res + ((long)n * C + c) * H * W : nullptr;
-  // 4-column register blocking: each thread owns a column quad; every
-  // K-row segment of the activation tile is loaded ONCE into registers
-  // and the 4 overlapping stencil windows read it there — (K+3)/4 ≈ 3 LDS
-  // reads per tap-row per output instead of K (the kernel was LDS-latency
-  // bound at 1 read per MAC)
+  // div-free pixel walk: software integer div/mod per element was ~8x the
+  // useful VALU work (SQ_INSTS_VALU 5075/wave vs ~650 needed)
   {
-    const int ncol4 = (W + 3) >> 2;
-    for (int t = threadIdx.x; t < tr * ncol4; t += CB2_BLOCK) {
-      const int r = t / ncol4, c0 = (t % ncol4) << 2;
-      float acc0 = b, acc1 = b, acc2 = b, acc3 = b;
+    int r = threadIdx.x / W, col = threadIdx.x % W;
+    const int dr = CB2_BLOCK / W, dc = CB2_BLOCK % W;
+    while (r < tr) {
+      float acc = b;
       #pragma unroll
-      for (int kr = 0; kr < K; ++kr) {  // K compile-time when KT > 0
-        const float* arow = a + (r + kr) * WP + c0;
-        float seg[CB2_MAX_K + 3];
+      for (int kr = 0; kr < K; ++kr) {
+        const float* arow = a + (r + kr) * WP + col;
         #pragma unroll
-        for (int i = 0; i < K + 3; ++i) seg[i] = arow[i];
-        #pragma unroll
-        for (int kc = 0; kc < K; ++kc) {
-          const float wv = ws[kr * K + kc];
-          acc0 += wv * seg[kc];
-          acc1 += wv * seg[kc + 1];
-          acc2 += wv * seg[kc + 2];
-          acc3 += wv * seg[kc + 3];
-        }
+        for (int kc = 0; kc < K; ++kc) acc += ws[kr * K + kc] * arow[kc];
       }
-      const float accs[4] = {acc0, acc1, acc2, acc3};
-      #pragma unroll
-      for (int j = 0; j < 4; ++j) {
-        const int col = c0 + j;
-        if (col < W) {
-          const long gp = (long)(r0 + r) * W + col;
-          float v = accs[j];
-          if (rpl != nullptr) v += ldf(rpl + gp);
-          stf(ypl + gp, v);
-        }
-      }
+      const long gp = (long)(r0 + r) * W + col;
+      if (rpl != nullptr) acc += ldf(rpl + gp);
+      stf(ypl + gp, acc);
+      r += dr;
+      col += dc;
+      if (col >= W) { col -= W; ++r; }
     }
   }
 }
@@ -152,50 +133,36 @@ npf_cb2d_bwd_dact(const T* __restrict__ x, const float* __restrict__ w,
   const int tr = min(CB2_TR, H - r0);
   T* dactpl = dact + ((long)n * C + c) * H * W;
   float s_dxhat = 0.f, s_dxhat_xhat = 0.f, s_dg = 0.f, s_db = 0.f, dbp = 0.f;
-  // 4-column register blocking (see the forward): each thread's 4
-  // transposed-stencil windows share a (K+3)-wide dY row segment loaded
-  // once into registers
   {
-    const int ncol4 = (W + 3) >> 2;
-    for (int t = threadIdx.x; t < tr * ncol4; t += CB2_BLOCK) {
-      const int r = t / ncol4, c0 = (t % ncol4) << 2;
-      float da0 = 0.f, da1 = 0.f, da2 = 0.f, da3 = 0.f;
+    int r = threadIdx.x / W, col = threadIdx.x % W;
+    const int drr = CB2_BLOCK / W, dcc = CB2_BLOCK % W;
+    while (r < tr) {
+      const long gp = (long)(r0 + r) * W + col;
+      const float dyl = ldf(dypl + gp);
+      dbp += dyl;
+      // transposed conv: da[p] = sum_k w[k] dY[p + pad - k]
+      float da = 0.f;
       #pragma unroll
       for (int kr = 0; kr < K; ++kr) {
-        // transposed conv: da[p] = sum_k w[k] dY[p + pad - k]
-        const float* drow = dys + (r + K - 1 - kr) * WP + c0;
-        float seg[CB2_MAX_K + 3];
+        const float* drow = dys + (r + K - 1 - kr) * WP + col;
         #pragma unroll
-        for (int i = 0; i < K + 3; ++i) seg[i] = drow[i];
-        #pragma unroll
-        for (int kc = 0; kc < K; ++kc) {
-          const float wv = ws[kr * K + kc];
-          da0 += wv * seg[K - 1 - kc];
-          da1 += wv * seg[K - 1 - kc + 1];
-          da2 += wv * seg[K - 1 - kc + 2];
-          da3 += wv * seg[K - 1 - kc + 3];
-        }
+        for (int kc = 0; kc < K; ++kc)
+          da += ws[kr * K + kc] * drow[K - 1 - kc];
       }
-      const float das[4] = {da0, da1, da2, da3};
-      #pragma unroll
-      for (int j = 0; j < 4; ++j) {
-        const int col = c0 + j;
-        if (col >= W) break;
-        const long gp = (long)(r0 + r) * W + col;
-        const float dyl = ldf(dypl + gp);
-        dbp += dyl;
-        const float act = a[(r + pad) * WP + col + pad];
-        const float dr = (act > 0.f) ? das[j] : 0.f;
-        stf(dactpl + gp, dr);
-        if (has_bn) {
-          const float xhat = (ldf(xpl + gp) - mu) * rs;
-          const float dxh = dr * gm;
-          s_dxhat += dxh;
-          s_dxhat_xhat += dxh * xhat;
-          s_dg += dr * xhat;
-          s_db += dr;
-        }
+      const float act = a[(r + pad) * WP + col + pad];
+      const float dr = (act > 0.f) ? da : 0.f;
+      stf(dactpl + gp, dr);
+      if (has_bn) {
+        const float xhat = (ldf(xpl + gp) - mu) * rs;
+        const float dxh = dr * gm;
+        s_dxhat += dxh;
+        s_dxhat_xhat += dxh * xhat;
+        s_dg += dr * xhat;
+        s_db += dr;
       }
+      r += drr;
+      col += dcc;
+      if (col >= W) { col -= W; ++r; }
     }
   }
   __syncthreads();

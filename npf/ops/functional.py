@@ -618,12 +618,12 @@ class _ConvBlock2dFn(torch.autograd.Function):
 
 
 def _cb2d_lds_ok(x, k):
-    """The 2D kernels stage 16-row tiles (+halo) in LDS; stay under the
-    64 KB default dynamic-LDS limit, else fall back to the composed path
-    (only reachable for images wider than ~330 px)."""
+    """The 2D kernels stage 8-row tiles (+halo) in LDS; stay under the 64 KB
+    default dynamic-LDS limit, else fall back to the composed path (only
+    reachable for images wider than ~780 px)."""
     W = x.shape[-1]
     pad = k // 2
-    return (2 * (16 + 2 * pad) * (W + 2 * pad) + k * k) * 4 <= 64 * 1024
+    return 2 * (8 + 2 * pad) * (W + 2 * pad) * 4 <= 64 * 1024
 
 
 def conv_block_2d(x, conv, bn=None, residual=None):

@@ -565,9 +565,7 @@ class TestAdviceGuards:
         assert _cb1d_lds_ok(ok, 19) and not _cb1d_lds_ok(big, 19)
         ok2 = torch.zeros(1, 8, 32, 32)
         big2 = torch.zeros(1, 8, 32, 2048)
-        mid2 = torch.zeros(1, 8, 32, 512)
         assert _cb2d_lds_ok(ok2, 9) and not _cb2d_lds_ok(big2, 9)
-        assert not _cb2d_lds_ok(mid2, 9)  # 16-row tiles cap W ~330
         assert _gde_lds_ok(ok2, 11) and not _gde_lds_ok(big2, 11)
 
     def test_mlp_chain_zero_rows_still_flows_grads(self):
