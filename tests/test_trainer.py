@@ -406,3 +406,44 @@ def test_device_episodes_image_grid():
     import math
 
     assert math.isfinite(tr.history[0]["train_loss"])
+
+
+def test_train_models_device_episodes_with_valid_split(tmp_path):
+    """train_models(device_episodes=...) with a CVSplit: the split happens
+    BEFORE wrapping (Subset of the raw dataset feeds DeviceEpisodes) and
+    validation keeps the collate path + best-valid checkpointing."""
+    from functools import partial
+
+    from npf import CNPFLoss
+    from npf.data.dataloader import cntxt_trgt_collate
+    from npf.data.imgs import SyntheticImages
+    from npf.train import CVSplit, train_models
+    from npf.utils.datasplit import GridCntxtTrgtGetter, RandomMasker, no_masker
+    from npf.zoo import gridconvcnp_2d
+
+    set_seed(0)
+    ds = SyntheticImages(shape=(3, 16, 16), n_samples=24)
+    test_ds = SyntheticImages(shape=(3, 16, 16), n_samples=8, split="test")
+    splitter = GridCntxtTrgtGetter(
+        context_masker=RandomMasker(a=0.05, b=0.2), target_masker=no_masker
+    )
+    collate = cntxt_trgt_collate(splitter, is_return_masks=True)
+    train_models(
+        {"synthetic": ds},
+        {"GridConvCNP": partial(gridconvcnp_2d, y_dim=3)},
+        CNPFLoss,
+        test_datasets={"synthetic": test_ds},
+        chckpnt_dirname=str(tmp_path) + "/",
+        is_retrain=True,
+        train_split=CVSplit(0.25),
+        max_epochs=1,
+        batch_size=8,
+        seed=0,
+        iterator_train__collate_fn=collate,
+        iterator_valid__collate_fn=collate,
+        device_episodes=partial(splitter, is_return_masks=True),
+    )
+    run = tmp_path / "synthetic" / "GridConvCNP" / "run_0"
+    assert (run / "params.pt").exists()
+    ll = np.loadtxt(run / "eval.csv", delimiter=",")
+    assert ll.shape == (8,) and np.isfinite(ll).all()
