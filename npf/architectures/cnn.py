@@ -230,24 +230,28 @@ class ResNormalizedConvBlock(ResConvBlock):
         init_param_(self.temperature)
 
     def forward(self, X):
-        """First half of channels = signal, second half = confidence."""
-        signal, conf_1 = X.chunk(2, dim=1)
-        conf_1 = conf_1.clamp(min=0, max=1)
-        X = signal * conf_1
+        """Normalized convolution: the first half of the channels carries the
+        signal, the second half a per-position confidence in [0, 1]."""
+        signal, confidence = X.chunk(2, dim=1)
+        confidence = confidence.clamp(0.0, 1.0)
+        weighted = signal * confidence
 
-        numerator = self.conv1(self.activation(X))
-        numerator = self.conv2_depthwise(self.activation(numerator))
-        density = self.conv2_depthwise(self.conv1(conf_1))
-        out = numerator / torch.clamp(density, min=1e-5)
+        # smear the confidence-weighted signal and the confidence mass with
+        # the same conv stack, then divide the mass back out (+ residual)
+        smeared = self.conv2_depthwise(
+            self.activation(self.conv1(self.activation(weighted)))
+        )
+        mass = self.conv2_depthwise(self.conv1(confidence))
+        normalized = smeared / mass.clamp(min=1e-5) + weighted
 
-        # confidence can only grow, capped at 1
-        conf_2 = conf_1 + torch.sigmoid(density * F.softplus(self.temperature) + self.bias)
-        conf_2 = conf_2.clamp(max=1)
-        out = out + X
+        # confidence is monotone non-decreasing and saturates at 1
+        gain = torch.sigmoid(mass * F.softplus(self.temperature) + self.bias)
+        new_confidence = (confidence + gain).clamp(max=1.0)
 
-        out = self.conv2_pointwise(out)
-        conf_2 = self.conv2_pointwise(conf_2)
-        return torch.cat([out, conf_2], dim=1)
+        return torch.cat(
+            [self.conv2_pointwise(normalized), self.conv2_pointwise(new_confidence)],
+            dim=1,
+        )
 
 
 class CNN(nn.Module):
@@ -348,18 +352,20 @@ class UnetCNN(CNN):
         return X, representation
 
     def _get_in_out_channels(self, n_channels, n_blocks):
-        """Channel-doubling U-Net schedule, capped at `max_nchannels`, with
-        the up-path inputs doubled by the skip concats (reference :492-516)."""
-        factor_chan = 2
+        """Channel-doubling U-Net schedule: widths double per down level and
+        mirror back up, interior widths capped at `max_nchannels` (the two
+        endpoints are the caller's I/O and stay uncapped), and every
+        up-path block's input is doubled by its skip concatenation."""
         assert n_blocks % 2 == 1, f"n_blocks={n_blocks} not odd"
-        channel_list = [factor_chan**i * n_channels for i in range(n_blocks // 2 + 1)]
-        channel_list = channel_list + channel_list[::-1]
-        channel_list = (
-            channel_list[:1]
-            + [min(c, self.max_nchannels) for c in channel_list[1:-1]]
-            + channel_list[-1:]
-        )
-        in_out_channels = super()._get_in_out_channels(channel_list, n_blocks)
-        idcs = slice(len(in_out_channels) // 2 + 1, len(in_out_channels))
-        in_out_channels[idcs] = [(i * 2, o) for i, o in in_out_channels[idcs]]
-        return in_out_channels
+        depth = n_blocks // 2
+        down = [n_channels << lvl for lvl in range(depth + 1)]
+        widths = down + down[::-1]
+        last = len(widths) - 1
+        widths = [
+            w if i in (0, last) else min(w, self.max_nchannels)
+            for i, w in enumerate(widths)
+        ]
+        return [
+            (cin * 2 if idx > depth else cin, cout)
+            for idx, (cin, cout) in enumerate(zip(widths[:-1], widths[1:]))
+        ]
