@@ -379,3 +379,68 @@ def test_gridconvcnp_xl_param_count():
 
     m = zoo.gridconvcnp_xl(y_dim=3)
     assert sum(p.numel() for p in m.parameters()) == 722417
+
+
+_REF_CNN_RUNNER = r"""
+import sys, warnings, torch
+warnings.filterwarnings("ignore")
+sys.path.insert(0, "/root/reference")
+from npf.architectures import CNN, UnetCNN, ResConvBlock
+from npf.architectures.cnn import ResNormalizedConvBlock
+import torch.nn as nn
+
+inp, outp = sys.argv[1], sys.argv[2]
+d = torch.load(inp, weights_only=False)
+
+# channel schedules for several Unet configs
+schedules = {}
+for n_ch, n_blocks, max_ch in [(16, 5, 256), (32, 7, 64), (8, 9, 32)]:
+    u = UnetCNN(n_ch, Conv=nn.Conv1d, ConvBlock=ResConvBlock, n_blocks=n_blocks,
+                kernel_size=5, max_nchannels=max_ch, Pool=nn.MaxPool1d,
+                upsample_mode="linear")
+    schedules[(n_ch, n_blocks, max_ch)] = u.in_out_channels
+
+# normalized-conv forward on fixed input/weights
+m = ResNormalizedConvBlock(8, 8, nn.Conv1d, kernel_size=5, n_conv_layers=2)
+m.load_state_dict(d["sd"])
+m.eval()
+with torch.no_grad():
+    y = m(d["x"])
+torch.save({"schedules": schedules, "y": y}, outp)
+"""
+
+
+@needs_ref
+def test_unet_schedule_and_normalized_block_match_reference(tmp_path):
+    """The rewritten UnetCNN channel schedule and ResNormalizedConvBlock
+    forward agree with the reference implementation exactly."""
+    import torch.nn as nn
+
+    from npf.architectures import CNN, ResConvBlock, UnetCNN
+    from npf.architectures.cnn import ResNormalizedConvBlock
+
+    torch.manual_seed(0)
+    ours_m = ResNormalizedConvBlock(8, 8, nn.Conv1d, kernel_size=5, n_conv_layers=2)
+    x = torch.rand(3, 16, 32)  # [B, 2*chan (signal;confidence), L]
+    inp, outp = str(tmp_path / "i.pt"), str(tmp_path / "o.pt")
+    torch.save({"sd": ours_m.state_dict(), "x": x}, inp)
+
+    env = dict(os.environ, PYTHONPATH=REF)
+    subprocess.run(
+        [sys.executable, "-c", _REF_CNN_RUNNER, inp, outp],
+        check=True, cwd="/tmp", env=env, capture_output=True,
+    )
+    ref = torch.load(outp, weights_only=False)
+
+    for (n_ch, n_blocks, max_ch), ref_sched in ref["schedules"].items():
+        u = UnetCNN(
+            n_ch, Conv=nn.Conv1d, ConvBlock=ResConvBlock, n_blocks=n_blocks,
+            kernel_size=5, max_nchannels=max_ch, Pool=nn.MaxPool1d,
+            upsample_mode="linear",
+        )
+        assert list(u.in_out_channels) == list(ref_sched), (n_ch, n_blocks, max_ch)
+
+    ours_m.eval()
+    with torch.no_grad():
+        y = ours_m(x)
+    assert torch.allclose(y, ref["y"], atol=1e-6), (y - ref["y"]).abs().max()
