@@ -316,13 +316,11 @@ class ZeroShotMultiMNIST(Dataset):
 
     def _transform(self, img_hw):
         if self.split == "train" and self.translation:
-            arr = (img_hw.numpy() * 255).astype(np.uint8)[..., None]
-            pad = self.translation
-            arr = np.pad(arr, [[pad, pad], [pad, pad], [0, 0]])
-            h0 = np.random.randint(0, 2 * pad + 1)
-            w0 = np.random.randint(0, 2 * pad + 1)
-            arr = arr[h0 : h0 + img_hw.shape[0], w0 : w0 + img_hw.shape[1], 0]
-            return torch.from_numpy(arr).float().unsqueeze(0) / 255.0
+            # reflect-padded random roll (reference random_translation,
+            # utils/data/helpers.py:157-171)
+            arr = np.atleast_3d(img_hw.numpy())
+            arr = random_translation(arr, self.translation)
+            return torch.from_numpy(arr[..., 0].copy()).float().unsqueeze(0)
         return img_hw.unsqueeze(0).float()
 
     def __getitem__(self, idx):

@@ -320,3 +320,25 @@ class TestSplitterOnDevice:
         Xg = torch.randn(2, 3, 16, 16, device="cuda")
         gout = GridCntxtTrgtGetter()(Xg)
         assert all(t.is_cuda for t in gout)
+
+
+def test_zsmm_translation_uses_reflect_roll():
+    """The zsmmt augmentation matches the reference's random_translation
+    semantics (reflect-padded roll, values preserved)."""
+    from npf.data.imgs import ZeroShotMultiMNIST, DatasetNotAvailable
+
+    try:
+        ds = ZeroShotMultiMNIST(split="train", translation=7)
+    except DatasetNotAvailable:
+        # no MNIST raw files in this environment: exercise _transform alone
+        ds = ZeroShotMultiMNIST.__new__(ZeroShotMultiMNIST)
+        ds.split = "train"
+        ds.translation = 7
+    torch.manual_seed(0)
+    img = torch.rand(56, 56)
+    out = ds._transform(img)
+    assert out.shape == (1, 56, 56)
+    # reflect-roll is value-preserving up to the rolled border region
+    assert out.min() >= 0 and out.max() <= 1
+    out2 = ds._transform(img)
+    assert not torch.equal(out, out2)  # random shift applied
